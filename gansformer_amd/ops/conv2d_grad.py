@@ -1,0 +1,131 @@
+"""conv2d with custom, infinitely differentiable autograd.
+
+Replaces the reference's cuDNN convs (TF 1.14 library kernels, SURVEY.md
+K3/L1) with our own MFMA implicit-GEMM HIP kernels on gfx950. Supports
+per-sample weights [B,O,I,kh,kw] natively — this is what makes modulated
+conv a single clean GEMM-shaped op instead of the grouped-conv trick the
+TF lineage used — plus shared weights [O,I,kh,kw], stride 1/2, square
+zero padding.
+
+Gradient structure (all pieces are themselves autograd Functions or plain
+differentiable torch ops, so R1 / path-length double-backward replays are
+exact):
+    dX = conv_fwd(zero_stuff(dY, stride), transpose_flip(W), stride=1)
+    dW = wgrad(X, dY)            (bilinear; its own Function)
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from . import native
+from .upfirdn2d import upfirdn2d
+
+
+def _eager_conv2d(x, w, stride, pad):
+    orig_dtype = x.dtype
+    if x.device.type == "cpu" and x.dtype not in (torch.float32, torch.float64):
+        x, w = x.float(), w.float()
+    if w.ndim == 4:
+        y = F.conv2d(x, w, stride=stride, padding=pad)
+    else:
+        B, O, I, kh, kw = w.shape
+        y = F.conv2d(
+            x.reshape(1, B * x.shape[1], *x.shape[2:]),
+            w.reshape(B * O, I, kh, kw),
+            stride=stride, padding=pad, groups=B)
+        y = y.reshape(B, O, *y.shape[2:])
+    return y.to(orig_dtype)
+
+
+def _eager_wgrad(x, dy, stride, pad, kh, kw, per_sample):
+    orig_dtype = x.dtype
+    if x.device.type == "cpu" and x.dtype not in (torch.float32, torch.float64):
+        x, dy = x.float(), dy.float()
+    B, I = x.shape[0], x.shape[1]
+    O = dy.shape[1]
+    unf = F.unfold(x, (kh, kw), padding=pad, stride=stride)  # [B, I*kh*kw, P]
+    dyf = dy.reshape(B, O, -1)  # [B, O, P]
+    dw = torch.einsum("bop,bip->boi", dyf, unf)  # [B, O, I*kh*kw]
+    dw = dw.reshape(B, O, I, kh, kw)
+    if not per_sample:
+        dw = dw.sum(0)
+    return dw.to(orig_dtype)
+
+
+def _transpose_flip(w):
+    """[.., O, I, kh, kw] -> [.., I, O, kh, kw] flipped spatially."""
+    return w.transpose(-4, -3).flip([-2, -1])
+
+
+def _id_filter(device):
+    return torch.ones(1, 1, dtype=torch.float32, device=device)
+
+
+def _conv_input_grad(dy, w, stride, pad, in_hw):
+    """dX for a forward conv with (w, stride, pad) and input size in_hw."""
+    kh, kw = w.shape[-2], w.shape[-1]
+    H, W = in_hw
+    Hout, Wout = dy.shape[-2], dy.shape[-1]
+    wt = _transpose_flip(w)
+    q0y, q0x = kh - 1 - pad, kw - 1 - pad
+    q1y = H + kh - 1 - Hout * stride - q0y
+    q1x = W + kw - 1 - Wout * stride - q0x
+    if stride == 1 and q0y == q1y and q0x == q1x and q0y >= 0 and q0y == q0x:
+        return conv2d_gradfix(dy, wt, stride=1, padding=q0y)
+    z = upfirdn2d(dy, _id_filter(dy.device), up=stride,
+                  padding=(q0y, q1y, q0x, q1x))
+    return conv2d_gradfix(z, wt, stride=1, padding=0)
+
+
+class _Conv2dWgrad(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, dy, stride, pad, kh, kw, per_sample):
+        ctx.save_for_backward(x, dy)
+        ctx.params = (stride, pad, kh, kw, per_sample)
+        if native.use_native(x, dy):
+            return native.require_ext().conv2d_wgrad(
+                x.contiguous(), dy.contiguous(), stride, pad, kh, kw, per_sample)
+        return _eager_wgrad(x, dy, stride, pad, kh, kw, per_sample)
+
+    @staticmethod
+    def backward(ctx, ddw):
+        x, dy = ctx.saved_tensors
+        stride, pad, kh, kw, per_sample = ctx.params
+        d_x = d_dy = None
+        if ctx.needs_input_grad[0]:
+            d_x = _conv_input_grad(dy, ddw, stride, pad, x.shape[-2:])
+        if ctx.needs_input_grad[1]:
+            d_dy = conv2d_gradfix(x, ddw, stride=stride, padding=pad)
+        return d_x, d_dy, None, None, None, None, None
+
+
+class _Conv2dFwd(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride, pad):
+        ctx.save_for_backward(x, w)
+        ctx.params = (stride, pad)
+        if native.use_native(x, w):
+            return native.require_ext().conv2d_fwd(
+                x.contiguous(), w.contiguous(), stride, pad)
+        return _eager_conv2d(x, w, stride, pad)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        stride, pad = ctx.params
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = _conv_input_grad(dy, w, stride, pad, x.shape[-2:])
+        if ctx.needs_input_grad[1]:
+            kh, kw = w.shape[-2], w.shape[-1]
+            dw = _Conv2dWgrad.apply(x, dy, stride, pad, kh, kw, w.ndim == 5)
+        return dx, dw, None, None
+
+
+def conv2d_gradfix(x, w, stride=1, padding=0):
+    """Conv2d (cross-correlation). x: [B,I,H,W]; w: [O,I,kh,kw] shared or
+    [B,O,I,kh,kw] per-sample. Returns [B,O,Ho,Wo]."""
+    assert x.ndim == 4 and w.ndim in (4, 5)
+    return _Conv2dFwd.apply(x, w, int(stride), int(padding))

@@ -1,0 +1,152 @@
+"""upfirdn2d: pad -> zero-stuff upsample -> 2D FIR filter -> downsample.
+
+Capability parity with the reference's `upfirdn_2d` CUDA op and its graph
+helpers (ref src/dnnlib/tflib/ops/upfirdn_2d.{py,cu} [R]):
+the blur behind every resample in G and D (default separable [1,3,3,1]
+filter). The gradient of upfirdn2d is upfirdn2d with the flipped filter
+and swapped up/down factors, so backward is implemented via the same
+autograd Function and the op is differentiable to all orders (needed for
+R1 / path-length second-order replays).
+
+Definition (per axis, up factor u, down factor d, pad (p0, p1)):
+    z[i*u] = x[i]                        (zero-stuffed, length W*u)
+    zp     = zero-pad z by (p0, p1)
+    y0[j]  = sum_t zp[j + t] * f[fw-1-t] (convolution with f, valid)
+    y      = y0[::d] * gain
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from . import native
+
+
+def setup_filter(f, device=None, normalize=True, gain=1.0):
+    """Build a 2D FIR filter tensor from a 1D (separable) or 2D spec."""
+    if f is None:
+        f = 1.0
+    f = torch.as_tensor(f, dtype=torch.float32)
+    if f.ndim == 0:
+        f = f[None]
+    if f.ndim == 1:
+        f = torch.outer(f, f)
+    assert f.ndim == 2
+    if normalize:
+        f = f / f.sum()
+    f = f * gain
+    if device is not None:
+        f = f.to(device)
+    return f
+
+
+def _eager_upfirdn2d(x, f, up, dn, pad, gain):
+    B, C, H, W = x.shape
+    uy, ux = up
+    dy, dx = dn
+    py0, py1, px0, px1 = pad
+    orig_dtype = x.dtype
+    if x.device.type == "cpu" and x.dtype not in (torch.float32, torch.float64):
+        x = x.float()
+    fh, fw = f.shape
+    # zero-stuff upsample
+    if uy > 1 or ux > 1:
+        z = x.new_zeros(B, C, H * uy, W * ux)
+        z[:, :, ::uy, ::ux] = x
+        x = z
+    # pad (negative pads crop)
+    x = F.pad(x, [px0, px1, py0, py1])
+    # convolution with f == correlation with flipped f
+    w = f.flip([0, 1]).to(x.dtype)[None, None].repeat(C, 1, 1, 1)
+    x = F.conv2d(x, w, groups=C)
+    # downsample
+    x = x[:, :, ::dy, ::dx]
+    if gain != 1.0:
+        x = x * gain
+    return x.to(orig_dtype)
+
+
+def _out_size(w, u, d, p0, p1, fw):
+    return (w * u + p0 + p1 - fw) // d + 1
+
+
+class _Upfirdn2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, f, up, dn, pad, gain):
+        ctx.params = (up, dn, pad, gain, x.shape)
+        ctx.save_for_backward(f)
+        if native.use_native(x):
+            return native.require_ext().upfirdn2d(
+                x.contiguous(), f.contiguous(),
+                up[1], up[0], dn[1], dn[0],
+                pad[2], pad[3], pad[0], pad[1], gain)
+        return _eager_upfirdn2d(x, f, up, dn, pad, gain)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (f,) = ctx.saved_tensors
+        up, dn, pad, gain, in_shape = ctx.params
+        fh, fw = f.shape
+        H, W = in_shape[2], in_shape[3]
+        Hout = _out_size(H, up[0], dn[0], pad[0], pad[1], fh)
+        Wout = _out_size(W, up[1], dn[1], pad[2], pad[3], fw)
+        # Adjoint: flipped filter, swapped up/down; pads solved so the
+        # output size equals the input size (derivation in module docstring
+        # coupling kernel: qy0 = fh-1-py0, qy1 from the size constraint).
+        qy0 = fh - 1 - pad[0]
+        qx0 = fw - 1 - pad[2]
+        qy1 = (H - 1) * up[0] + fh - qy0 - Hout * dn[0]
+        qx1 = (W - 1) * up[1] + fw - qx0 - Wout * dn[1]
+        dx = _Upfirdn2d.apply(dy, f.flip([0, 1]), dn, up, (qy0, qy1, qx0, qx1), gain)
+        return dx, None, None, None, None, None
+
+
+def _pair(v):
+    if isinstance(v, (tuple, list)):
+        return tuple(int(t) for t in v)
+    return (int(v), int(v))
+
+
+def upfirdn2d(x, f, up=1, down=1, padding=0, gain=1.0):
+    """x: [B,C,H,W]; f: 2D filter (see setup_filter).
+
+    up/down: int or (y, x). padding: int, (x, y) pair applied to both
+    sides, or explicit (py0, py1, px0, px1).
+    """
+    assert x.ndim == 4
+    up = _pair(up)
+    down = _pair(down)
+    if isinstance(padding, (tuple, list)) and len(padding) == 4:
+        pad = tuple(int(p) for p in padding)
+    elif isinstance(padding, (tuple, list)) and len(padding) == 2:
+        py, px = int(padding[0]), int(padding[1])
+        pad = (py, py, px, px)
+    else:
+        p = int(padding)
+        pad = (p, p, p, p)
+    if not isinstance(f, torch.Tensor):
+        f = setup_filter(f, device=x.device)
+    f = f.to(device=x.device, dtype=torch.float32)
+    return _Upfirdn2d.apply(x, f, up, down, pad, float(gain))
+
+
+def upsample2d(x, f, up=2, gain=1.0):
+    fh, fw = f.shape if isinstance(f, torch.Tensor) else (len(f), len(f))
+    p0 = (fh + up - 1) // 2
+    p1 = (fh - up) // 2
+    return upfirdn2d(x, f, up=up, padding=(p0, p1, p0, p1), gain=gain * up * up)
+
+
+def downsample2d(x, f, down=2, gain=1.0):
+    fh, fw = f.shape if isinstance(f, torch.Tensor) else (len(f), len(f))
+    p0 = (fh - down + 1) // 2
+    p1 = (fh - down) // 2
+    return upfirdn2d(x, f, down=down, padding=(p0, p1, p0, p1), gain=gain)
+
+
+def filter2d(x, f, gain=1.0):
+    fh, fw = f.shape if isinstance(f, torch.Tensor) else (len(f), len(f))
+    p0 = (fh - 1) // 2
+    p1 = fh // 2
+    return upfirdn2d(x, f, padding=(p0, p1, p0, p1), gain=gain)

@@ -1,0 +1,158 @@
+"""CPU (eager-reference) correctness of the op layer: forward semantics,
+first- and second-order gradients. These eager paths are the golden
+references the GPU kernel tests (test_gpu_ops.py) compare against."""
+
+import math
+
+import pytest
+import torch
+from torch.autograd import gradcheck, gradgradcheck
+
+from gansformer_amd.ops import (bias_act, bipartite_attention,
+                                conv2d_gradfix, minibatch_stddev,
+                                modulated_conv2d, setup_filter, upfirdn2d,
+                                upsample2d, downsample2d)
+
+
+def test_bias_act_forward_lrelu():
+    x = torch.randn(2, 4, 5, 5)
+    b = torch.randn(4)
+    y = bias_act(x, b, act="lrelu")
+    ref = torch.nn.functional.leaky_relu(x + b.view(1, -1, 1, 1), 0.2) * math.sqrt(2)
+    assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_bias_act_clamp():
+    x = torch.randn(8, 3) * 10
+    y = bias_act(x.unsqueeze(-1), None, act="linear", clamp=0.5)
+    assert y.abs().max() <= 0.5 + 1e-6
+
+
+@pytest.mark.parametrize("act", ["linear", "relu", "lrelu"])
+def test_bias_act_gradcheck(act):
+    x = torch.randn(3, 4, 2, 2, dtype=torch.float64, requires_grad=True)
+    b = torch.randn(4, dtype=torch.float64, requires_grad=True)
+    # keep away from the kink for finite differences
+    x = (x + 0.5 * torch.sign(x)).detach().requires_grad_(True)
+    assert gradcheck(lambda x_, b_: bias_act(x_, b_, act=act), (x, b),
+                     eps=1e-6, atol=1e-4)
+    assert gradgradcheck(lambda x_, b_: bias_act(x_, b_, act=act), (x, b),
+                         eps=1e-6, atol=1e-4)
+
+
+@pytest.mark.parametrize("up,down", [(1, 1), (2, 1), (1, 2), (2, 2)])
+def test_upfirdn2d_gradcheck(up, down):
+    f = setup_filter([1, 3, 3, 1]).to(torch.float64)
+    x = torch.randn(2, 3, 8, 8, dtype=torch.float64, requires_grad=True)
+    fn = lambda x_: upfirdn2d(x_, f, up=up, down=down, padding=2)
+    assert gradcheck(fn, (x,), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(fn, (x,), eps=1e-6, atol=1e-4)
+
+
+def test_upsample_downsample_shapes():
+    f = setup_filter([1, 3, 3, 1])
+    x = torch.randn(1, 2, 16, 16)
+    assert upsample2d(x, f).shape == (1, 2, 32, 32)
+    assert downsample2d(x, f).shape == (1, 2, 8, 8)
+
+
+def test_upsample_preserves_dc():
+    """Upsampling a constant image with a normalized filter keeps the level."""
+    f = setup_filter([1, 3, 3, 1])
+    x = torch.ones(1, 1, 8, 8)
+    y = upsample2d(x, f)
+    # interior pixels should be ~1.0
+    assert torch.allclose(y[:, :, 4:-4, 4:-4], torch.ones(1), atol=1e-5)
+
+
+@pytest.mark.parametrize("per_sample", [False, True])
+@pytest.mark.parametrize("stride,k", [(1, 3), (1, 1), (2, 3)])
+def test_conv2d_gradfix_matches_torch(per_sample, stride, k):
+    B, I, O, H = 2, 3, 4, 8
+    x = torch.randn(B, I, H, H)
+    if per_sample:
+        w = torch.randn(B, O, I, k, k)
+        ref = torch.nn.functional.conv2d(
+            x.reshape(1, B * I, H, H), w.reshape(B * O, I, k, k),
+            stride=stride, padding=k // 2, groups=B).reshape(B, O, -1)
+        y = conv2d_gradfix(x, w, stride=stride, padding=k // 2)
+        assert torch.allclose(y.reshape(B, O, -1), ref, atol=1e-5)
+    else:
+        w = torch.randn(O, I, k, k)
+        ref = torch.nn.functional.conv2d(x, w, stride=stride, padding=k // 2)
+        y = conv2d_gradfix(x, w, stride=stride, padding=k // 2)
+        assert torch.allclose(y, ref, atol=1e-5)
+
+
+@pytest.mark.parametrize("per_sample", [False, True])
+@pytest.mark.parametrize("stride", [1, 2])
+def test_conv2d_gradfix_gradcheck(per_sample, stride):
+    B, I, O, H, k = 2, 2, 3, 6, 3
+    x = torch.randn(B, I, H, H, dtype=torch.float64, requires_grad=True)
+    wshape = (B, O, I, k, k) if per_sample else (O, I, k, k)
+    w = torch.randn(*wshape, dtype=torch.float64, requires_grad=True)
+    fn = lambda x_, w_: conv2d_gradfix(x_, w_, stride=stride, padding=1)
+    assert gradcheck(fn, (x, w), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(fn, (x, w), eps=1e-6, atol=1e-4)
+
+
+def test_modulated_conv_demod_unit_variance():
+    """With demodulation, output std should be ~1 for unit-variance input."""
+    torch.manual_seed(0)
+    B, I, O, H = 4, 64, 64, 16
+    x = torch.randn(B, I, H, H)
+    w = torch.randn(O, I, 3, 3)
+    s = torch.randn(B, I).exp()
+    y = modulated_conv2d(x, w, s, demodulate=True)
+    assert y.shape == (B, O, H, H)
+    assert 0.7 < y.std().item() < 1.4
+
+
+def test_modulated_conv_up():
+    f = setup_filter([1, 3, 3, 1])
+    x = torch.randn(2, 8, 8, 8)
+    w = torch.randn(4, 8, 3, 3)
+    s = torch.randn(2, 8)
+    y = modulated_conv2d(x, w, s, up=2, resample_filter=f)
+    assert y.shape == (2, 4, 16, 16)
+
+
+def test_modulated_conv_down():
+    f = setup_filter([1, 3, 3, 1])
+    x = torch.randn(2, 8, 16, 16)
+    w = torch.randn(4, 8, 3, 3)
+    s = torch.randn(2, 8)
+    y = modulated_conv2d(x, w, s, down=2, resample_filter=f)
+    assert y.shape == (2, 4, 8, 8)
+
+
+def test_minibatch_stddev():
+    x = torch.randn(4, 8, 4, 4)
+    y = minibatch_stddev(x, group_size=2, num_channels=1)
+    assert y.shape == (4, 9, 4, 4)
+    # identical samples in a group -> zero stddev feature
+    x2 = x.clone()
+    x2[2:] = x2[:2]  # groups are strided: b = g*(B//G)+m, G=2 -> pairs (0,2),(1,3)
+    y2 = minibatch_stddev(x2, group_size=2, num_channels=1)
+    assert y2[:, 8].abs().max() < 1e-3
+
+
+def test_bipartite_attention_matches_manual():
+    torch.manual_seed(1)
+    q = torch.randn(2, 10, 8)
+    k = torch.randn(2, 4, 8)
+    v = torch.randn(2, 4, 6)
+    out = bipartite_attention(q, k, v)
+    a = torch.softmax(q @ k.transpose(1, 2) / math.sqrt(8), dim=-1)
+    ref = a @ v
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_bipartite_attention_gradcheck():
+    q = torch.randn(1, 5, 4, dtype=torch.float64, requires_grad=True)
+    k = torch.randn(1, 3, 4, dtype=torch.float64, requires_grad=True)
+    v = torch.randn(1, 3, 4, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda *t: bipartite_attention(*t), (q, k, v),
+                     eps=1e-6, atol=1e-4)
+    assert gradgradcheck(lambda *t: bipartite_attention(*t), (q, k, v),
+                         eps=1e-6, atol=1e-4)
