@@ -1,0 +1,83 @@
+// Common helpers for the gansformer_amd gfx950 HIP kernels.
+// Target: MI355X (CDNA4, wave64, MFMA). No CUDA compatibility paths.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define GFA_DEV __device__ __forceinline__
+
+namespace gfa {
+
+using bf16 = __hip_bfloat16;
+
+typedef float  f32x4  __attribute__((ext_vector_type(4)));
+typedef float  f32x16 __attribute__((ext_vector_type(16)));
+typedef short  s16x4  __attribute__((ext_vector_type(4)));
+typedef short  s16x8  __attribute__((ext_vector_type(8)));
+typedef unsigned short u16;
+
+// ---- scalar dtype conversion ----
+GFA_DEV float to_f32(float v) { return v; }
+GFA_DEV float to_f32(bf16 v) { return __bfloat162float(v); }
+GFA_DEV float bf16_bits_to_f32(u16 bits) {
+  union { unsigned int u; float f; } cvt;
+  cvt.u = (unsigned int)bits << 16;
+  return cvt.f;
+}
+GFA_DEV u16 f32_to_bf16_bits(float v) {
+  // round-to-nearest-even
+  union { float f; unsigned int u; } cvt;
+  cvt.f = v;
+  unsigned int lsb = (cvt.u >> 16) & 1u;
+  unsigned int rounded = cvt.u + 0x7FFFu + lsb;
+  return (u16)(rounded >> 16);
+}
+
+template <typename T> GFA_DEV T from_f32(float v);
+template <> GFA_DEV float from_f32<float>(float v) { return v; }
+template <> GFA_DEV bf16 from_f32<bf16>(float v) {
+  return __float2bfloat16(v);
+}
+
+// raw (bit-level) storage type: __hip_bfloat16 is a struct and cannot be
+// an ext_vector element, so vectorized kernels work on u16 bits.
+template <typename T> struct Raw { using type = T; };
+template <> struct Raw<bf16> { using type = u16; };
+
+GFA_DEV float raw_to_f32(float v) { return v; }
+GFA_DEV float raw_to_f32(u16 v) { return bf16_bits_to_f32(v); }
+template <typename R> GFA_DEV R f32_to_raw(float v);
+template <> GFA_DEV float f32_to_raw<float>(float v) { return v; }
+template <> GFA_DEV u16 f32_to_raw<u16>(float v) {
+  return f32_to_bf16_bits(v);
+}
+
+// ---- grid helpers ----
+GFA_DEV long global_tid() {
+  return (long)blockIdx.x * blockDim.x + threadIdx.x;
+}
+GFA_DEV long global_stride() {
+  return (long)gridDim.x * blockDim.x;
+}
+
+inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+// memory-bound launch: cap grid, grid-stride the rest (guide G11)
+inline dim3 stream_grid(long n, int block = 256, int max_blocks = 2048) {
+  long blocks = (n + block - 1) / block;
+  if (blocks > max_blocks) blocks = max_blocks;
+  if (blocks < 1) blocks = 1;
+  return dim3((unsigned)blocks);
+}
+
+}  // namespace gfa
+
+#define HIP_CHECK_LAST()                                                    \
+  do {                                                                      \
+    hipError_t _e = hipGetLastError();                                      \
+    if (_e != hipSuccess) {                                                 \
+      TORCH_CHECK(false, "HIP kernel launch failed: ",                      \
+                  hipGetErrorString(_e));                                   \
+    }                                                                       \
+  } while (0)

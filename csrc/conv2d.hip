@@ -1,0 +1,343 @@
+// Implicit-GEMM conv2d on MFMA for gfx950, with native per-sample weights.
+// Replaces the reference's cuDNN convs + grouped-conv modulation trick
+// (ref src/training/networks.py modulated conv).
+//
+// Forward GEMM view (per sample b):
+//   A = W[b?]           [O][K]   K = I*kh*kw   (contiguous rows)
+//   B = X patches       [K][P]   implicit im2col addressing
+//   C = Y[b]            [O][P]
+// Tile: 64(O) x 64(P) x BK, 4 waves (2x2), each wave a 32x32 sub-tile of
+// 2x2 MFMA fragments. A and B~ tiles staged in LDS with an 8-element XOR
+// swizzle so the 16-byte fragment reads are bank-spread (guide T2).
+//
+// Weight-gradient GEMM view (wgrad):
+//   A = dY[b]           [O][P]
+//   B = X patches       [Kw][P]  Kw = I*kh*kw
+//   C = dW[b?]          [O][Kw]  (shared weights: sum over b in the K loop)
+//
+// bf16 path: v_mfma_f32_16x16x32_bf16.  f32 path: v_mfma_f32_16x16x4_f32
+// (exact f32 at the f32 vector rate — there is no xf32 on gfx950).
+#include "common.h"
+
+namespace gfa {
+
+struct ConvParams {
+  int B, I, H, W;       // input
+  int O, OH, OW;        // output
+  int kh, kw, stride, pad;
+  int per_sample;       // weights have a leading B dim
+};
+
+// ---------------- bf16 forward ----------------
+// LDS tiles: As[64][32] bf16 (A rows), Bs[64][32] bf16 (pixel rows, k cols)
+// swizzle: 8-element granule g at row r stored at g ^ (r & 3).
+
+__device__ __forceinline__ int swz(int row, int g) { return g ^ (row & 3); }
+
+__global__ __launch_bounds__(256)
+void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
+                     const bf16* __restrict__ w, ConvParams p) {
+  constexpr int BM = 64, BN = 64, BK = 32;
+  __shared__ u16 As[BM * BK];
+  __shared__ u16 Bs[BN * BK];
+
+  const int K = p.I * p.kh * p.kw;
+  const int P = p.OH * p.OW;
+  const int b = blockIdx.z;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  const bf16* wb = w + (p.per_sample ? (long)b * p.O * K : 0);
+  const bf16* xb = x + (long)b * p.I * p.H * p.W;
+  bf16* yb = y + (long)b * p.O * P;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 32;  // wave row offset in tile
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2] = {};
+  const bool a_vec = (K % 8 == 0);
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A: 64 rows x 32 k, thread t -> row t/4, granule t%4
+    {
+      int row = t >> 2, g = t & 3;
+      int o = m0 + row;
+      int kk = k0 + g * 8;
+      u16* dst = &As[row * BK + swz(row, g) * 8];
+      if (o < p.O && kk + 8 <= K && a_vec) {
+        *reinterpret_cast<s16x8*>(dst) =
+            *reinterpret_cast<const s16x8*>(wb + (long)o * K + kk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16 v = 0;
+          if (o < p.O && kk + j < K)
+            v = __builtin_bit_cast(u16, wb[(long)o * K + kk + j]);
+          dst[j] = v;
+        }
+      }
+    }
+    // ---- stage B: 64 pixels x 32 k; pass over k-slices, coalesced in p
+#pragma unroll
+    for (int pass = 0; pass < 8; ++pass) {
+      int kk = k0 + pass * 4 + (t >> 6);  // 4 k per pass x 4 waves
+      int pix = n0 + (t & 63);
+      u16 v = 0;
+      if (kk < K && pix < P) {
+        int i = kk / (p.kh * p.kw);
+        int rs = kk % (p.kh * p.kw);
+        int r = rs / p.kw, s = rs % p.kw;
+        int oy = pix / p.OW, ox = pix % p.OW;
+        int iy = oy * p.stride + r - p.pad;
+        int ix = ox * p.stride + s - p.pad;
+        if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
+          v = __builtin_bit_cast(u16, xb[((long)i * p.H + iy) * p.W + ix]);
+      }
+      int lk = kk - k0;
+      int lp = pix - n0;
+      Bs[lp * BK + (swz(lp, lk >> 3) << 3) + (lk & 7)] = v;
+    }
+    __syncthreads();
+    // ---- MFMA: each wave 2x2 fragments of 16x16, K=32 in one mfma each
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int arow = wm + mi * 16 + (lane & 15);
+        int ag = lane >> 4;
+        s16x8 af = *reinterpret_cast<const s16x8*>(
+            &As[arow * BK + swz(arow, ag) * 8]);
+        int brow = wn + ni * 16 + (lane & 15);
+        s16x8 bf = *reinterpret_cast<const s16x8*>(
+            &Bs[brow * BK + swz(brow, ag) * 8]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  // ---- epilogue: C/D map col=lane&15, row=(lane>>4)*4+reg
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = n0 + wn + ni * 16 + (lane & 15);
+      if (col >= P) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        if (row < p.O)
+          yb[(long)row * P + col] = from_f32<bf16>(acc[mi][ni][reg]);
+      }
+    }
+}
+
+// ---------------- f32 forward (exact f32, 16x16x4 MFMA) ----------------
+__global__ __launch_bounds__(256)
+void conv2d_fwd_f32(float* __restrict__ y, const float* __restrict__ x,
+                    const float* __restrict__ w, ConvParams p) {
+  constexpr int BM = 64, BN = 64, BK = 16;
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BN][BK + 1];
+
+  const int K = p.I * p.kh * p.kw;
+  const int P = p.OH * p.OW;
+  const int b = blockIdx.z;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  const float* wb = w + (p.per_sample ? (long)b * p.O * K : 0);
+  const float* xb = x + (long)b * p.I * p.H * p.W;
+  float* yb = y + (long)b * p.O * P;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {  // stage A: thread t -> row t/4, k-chunk of 4
+      int row = t >> 2, c4 = (t & 3) * 4;
+      int o = m0 + row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int kk = k0 + c4 + j;
+        As[row][c4 + j] =
+            (o < p.O && kk < K) ? wb[(long)o * K + kk] : 0.f;
+      }
+    }
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      int kk = k0 + pass * 4 + (t >> 6);
+      int pix = n0 + (t & 63);
+      float v = 0.f;
+      if (kk < K && pix < P) {
+        int i = kk / (p.kh * p.kw);
+        int rs = kk % (p.kh * p.kw);
+        int r = rs / p.kw, s = rs % p.kw;
+        int oy = pix / p.OW, ox = pix % p.OW;
+        int iy = oy * p.stride + r - p.pad;
+        int ix = ox * p.stride + s - p.pad;
+        if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
+          v = xb[((long)i * p.H + iy) * p.W + ix];
+      }
+      Bs[pix - n0][kk - k0] = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          float av = As[wm + mi * 16 + (lane & 15)][kk + (lane >> 4)];
+          float bv = Bs[wn + ni * 16 + (lane & 15)][kk + (lane >> 4)];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              av, bv, acc[mi][ni], 0, 0, 0);
+        }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = n0 + wn + ni * 16 + (lane & 15);
+      if (col >= P) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        if (row < p.O) yb[(long)row * P + col] = acc[mi][ni][reg];
+      }
+    }
+}
+
+// ---------------- wgrad ----------------
+// dW[b?][o][kw_idx] = sum_p dY[b][o][p] * patch[b][kw_idx][p]
+// A rows: dY (contiguous in p). B rows: implicit patches (contiguous-ish).
+// Shared weights: K loop covers (b, p); per-sample: fixed b per block.
+template <typename T>
+__global__ __launch_bounds__(256)
+void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
+                         const T* __restrict__ dy, ConvParams p) {
+  constexpr int BM = 64, BN = 64, BK = 32;
+  __shared__ float As[BM][BK / 8][9];   // [row][granule][8+1 pad] as floats
+  __shared__ float Bs[BN][BK / 8][9];
+
+  const int Kw = p.I * p.kh * p.kw;
+  const int P = p.OH * p.OW;
+  const int m0 = blockIdx.y * BM;      // O tile
+  const int n0 = blockIdx.x * BN;      // Kw tile
+  const int nb = p.per_sample ? 1 : p.B;
+  const int b_fix = p.per_sample ? blockIdx.z : 0;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2] = {};
+
+  for (int bb = 0; bb < nb; ++bb) {
+    const int b = p.per_sample ? b_fix : bb;
+    const T* dyb = dy + (long)b * p.O * P;
+    const T* xb = x + (long)b * p.I * p.H * p.W;
+    for (int p0 = 0; p0 < P; p0 += BK) {
+      {  // stage A (dY): row t/4, granule t%4 -> 8 p values
+        int row = t >> 2, g = t & 3;
+        int o = m0 + row;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int pp = p0 + g * 8 + j;
+          As[row][g][j] = (o < p.O && pp < P)
+                              ? to_f32(dyb[(long)o * P + pp]) : 0.f;
+        }
+      }
+      {  // stage B (patches): row = Kw index, same granule layout
+        int row = t >> 2, g = t & 3;
+        int kwi = n0 + row;
+        int i = kwi / (p.kh * p.kw);
+        int rs = kwi % (p.kh * p.kw);
+        int r = rs / p.kw, s = rs % p.kw;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int pp = p0 + g * 8 + j;
+          float v = 0.f;
+          if (kwi < Kw && pp < P) {
+            int oy = pp / p.OW, ox = pp % p.OW;
+            int iy = oy * p.stride + r - p.pad;
+            int ix = ox * p.stride + s - p.pad;
+            if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
+              v = to_f32(xb[((long)i * p.H + iy) * p.W + ix]);
+          }
+          Bs[row][g][j] = v;
+        }
+      }
+      __syncthreads();
+      // f32 MFMA over the staged K-slab (values may be bf16-origin; the
+      // products are exact in f32)
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+#pragma unroll
+        for (int kk = 0; kk < 8; kk += 4) {
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < 2; ++ni) {
+              float av = As[wm + mi * 16 + (lane & 15)][g][kk + (lane >> 4)];
+              float bv = Bs[wn + ni * 16 + (lane & 15)][g][kk + (lane >> 4)];
+              acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                  av, bv, acc[mi][ni], 0, 0, 0);
+            }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  const int Kw2 = Kw;
+  T* dwb = dw + (p.per_sample ? (long)b_fix * p.O * Kw2 : 0);
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = n0 + wn + ni * 16 + (lane & 15);
+      if (col >= Kw2) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        if (row < p.O)
+          dwb[(long)row * Kw2 + col] = from_f32<T>(acc[mi][ni][reg]);
+      }
+    }
+}
+
+// ---------------- launchers ----------------
+void launch_conv2d_fwd_bf16(bf16* y, const bf16* x, const bf16* w,
+                            const ConvParams& p, hipStream_t s) {
+  dim3 grid(ceil_div(p.OH * (long)p.OW, 64), ceil_div(p.O, 64), p.B);
+  hipLaunchKernelGGL(conv2d_fwd_bf16, grid, dim3(256), 0, s, y, x, w, p);
+}
+void launch_conv2d_fwd_f32(float* y, const float* x, const float* w,
+                           const ConvParams& p, hipStream_t s) {
+  dim3 grid(ceil_div(p.OH * (long)p.OW, 64), ceil_div(p.O, 64), p.B);
+  hipLaunchKernelGGL(conv2d_fwd_f32, grid, dim3(256), 0, s, y, x, w, p);
+}
+template <typename T>
+void launch_conv2d_wgrad(T* dw, const T* x, const T* dy, const ConvParams& p,
+                         hipStream_t s) {
+  int Kw = p.I * p.kh * p.kw;
+  dim3 grid(ceil_div(Kw, 64), ceil_div(p.O, 64), p.per_sample ? p.B : 1);
+  hipLaunchKernelGGL(conv2d_wgrad_kernel<T>, grid, dim3(256), 0, s, dw, x,
+                     dy, p);
+}
+template void launch_conv2d_wgrad<float>(float*, const float*, const float*,
+                                         const ConvParams&, hipStream_t);
+template void launch_conv2d_wgrad<bf16>(bf16*, const bf16*, const bf16*,
+                                        const ConvParams&, hipStream_t);
+
+}  // namespace gfa

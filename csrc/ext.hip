@@ -1,0 +1,274 @@
+// Torch bindings for the gansformer_amd gfx950 HIP kernels.
+// Replaces the reference's import-time nvcc JIT + tf.load_op_library
+// (ref src/dnnlib/tflib/custom_ops.py [R]) with an AOT-built extension.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace gfa {
+
+struct UfdParams {
+  int B, C, H, W;
+  int OH, OW;
+  int fh, fw;
+  int upx, upy, downx, downy;
+  int px0, px1, py0, py1;
+  float gain;
+};
+
+struct ConvParams {
+  int B, I, H, W;
+  int O, OH, OW;
+  int kh, kw, stride, pad;
+  int per_sample;
+};
+
+struct AttnParams {
+  int B, Nq, Nk, D, E;
+  float scale;
+};
+
+template <typename T>
+void launch_fba(T*, const T*, const float*, const T*, long, long, int, int,
+                int, float, float, float, hipStream_t);
+template <typename T>
+void launch_upfirdn2d(T*, const T*, const float*, const UfdParams&,
+                      hipStream_t);
+void launch_conv2d_fwd_bf16(bf16*, const bf16*, const bf16*,
+                            const ConvParams&, hipStream_t);
+void launch_conv2d_fwd_f32(float*, const float*, const float*,
+                           const ConvParams&, hipStream_t);
+template <typename T>
+void launch_conv2d_wgrad(T*, const T*, const T*, const ConvParams&,
+                         hipStream_t);
+template <typename T>
+void launch_mbstd(float*, const T*, int, int, int, int, int, float,
+                  hipStream_t);
+template <typename T>
+void launch_attn_smalln(T*, const T*, const T*, const T*, const AttnParams&,
+                        hipStream_t);
+template <typename T>
+void launch_attn_longn(T*, const T*, const T*, const T*, float*, float*,
+                       float*, const AttnParams&, int, hipStream_t);
+
+}  // namespace gfa
+
+namespace {
+
+using torch::Tensor;
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_dtype(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32 ||
+                  t.scalar_type() == torch::kBFloat16,
+              name, ": expected float32 or bfloat16, got ", t.scalar_type());
+}
+
+#define DISPATCH_FT(t, NAME, ...)                                         \
+  do {                                                                    \
+    if ((t).scalar_type() == torch::kFloat32) {                           \
+      using scalar_t = float;                                             \
+      __VA_ARGS__;                                                        \
+    } else {                                                              \
+      using scalar_t = gfa::bf16;                                         \
+      __VA_ARGS__;                                                        \
+    }                                                                     \
+  } while (0)
+
+template <typename T>
+T* ptr(Tensor& t) { return reinterpret_cast<T*>(t.data_ptr()); }
+template <typename T>
+const T* cptr(const Tensor& t) { return reinterpret_cast<const T*>(t.data_ptr()); }
+
+Tensor fba(Tensor x, Tensor b, Tensor ref, int64_t act, int64_t grad,
+           double alpha, double gain, double clamp) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  check_dtype(x, "fba.x");
+  auto out = torch::empty_like(x);
+  long n = x.numel();
+  long inner = 1;
+  int C = 1;
+  Tensor b32;
+  const float* bptr = nullptr;
+  if (grad == 0 && b.numel() > 0) {
+    TORCH_CHECK(x.dim() >= 2, "fba: bias needs a channel dim");
+    C = (int)x.size(1);
+    for (int d = 2; d < x.dim(); ++d) inner *= x.size(d);
+    b32 = b.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(b32.numel() == C, "fba: bias size mismatch");
+    bptr = b32.data_ptr<float>();
+  }
+  if (grad == 1) {
+    TORCH_CHECK(ref.sizes() == x.sizes() && ref.scalar_type() == x.scalar_type(),
+                "fba: ref must match x");
+  }
+  DISPATCH_FT(x, "fba", {
+    gfa::launch_fba<scalar_t>(
+        ptr<scalar_t>(out), cptr<scalar_t>(x), bptr,
+        grad == 1 ? cptr<scalar_t>(ref) : nullptr, n, inner, C, (int)act,
+        (int)grad, (float)alpha, (float)gain, (float)clamp, cur_stream());
+  });
+  return out;
+}
+
+Tensor upfirdn2d(Tensor x, Tensor f, int64_t upx, int64_t upy, int64_t downx,
+                 int64_t downy, int64_t px0, int64_t px1, int64_t py0,
+                 int64_t py1, double gain) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(f.is_cuda() && f.is_contiguous() && f.dim() == 2 &&
+              f.scalar_type() == torch::kFloat32);
+  check_dtype(x, "upfirdn2d.x");
+  gfa::UfdParams p;
+  p.B = (int)x.size(0); p.C = (int)x.size(1);
+  p.H = (int)x.size(2); p.W = (int)x.size(3);
+  p.fh = (int)f.size(0); p.fw = (int)f.size(1);
+  p.upx = (int)upx; p.upy = (int)upy;
+  p.downx = (int)downx; p.downy = (int)downy;
+  p.px0 = (int)px0; p.px1 = (int)px1; p.py0 = (int)py0; p.py1 = (int)py1;
+  p.gain = (float)gain;
+  p.OH = (int)((p.H * p.upy + p.py0 + p.py1 - p.fh) / p.downy + 1);
+  p.OW = (int)((p.W * p.upx + p.px0 + p.px1 - p.fw) / p.downx + 1);
+  TORCH_CHECK(p.OH > 0 && p.OW > 0, "upfirdn2d: empty output");
+  auto out = torch::empty({p.B, p.C, p.OH, p.OW}, x.options());
+  DISPATCH_FT(x, "upfirdn2d", {
+    gfa::launch_upfirdn2d<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(x),
+                                    f.data_ptr<float>(), p, cur_stream());
+  });
+  return out;
+}
+
+gfa::ConvParams conv_params(const Tensor& x, int O, int kh, int kw,
+                            int stride, int pad, bool per_sample) {
+  gfa::ConvParams p;
+  p.B = (int)x.size(0); p.I = (int)x.size(1);
+  p.H = (int)x.size(2); p.W = (int)x.size(3);
+  p.O = O; p.kh = kh; p.kw = kw; p.stride = stride; p.pad = pad;
+  p.per_sample = per_sample ? 1 : 0;
+  p.OH = (p.H + 2 * pad - kh) / stride + 1;
+  p.OW = (p.W + 2 * pad - kw) / stride + 1;
+  TORCH_CHECK(p.OH > 0 && p.OW > 0, "conv2d: empty output");
+  return p;
+}
+
+Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous());
+  TORCH_CHECK(w.dim() == 4 || w.dim() == 5, "conv2d: w must be 4D or 5D");
+  TORCH_CHECK(x.scalar_type() == w.scalar_type(), "conv2d: dtype mismatch");
+  check_dtype(x, "conv2d.x");
+  bool per_sample = (w.dim() == 5);
+  if (per_sample) TORCH_CHECK(w.size(0) == x.size(0), "conv2d: batch mismatch");
+  int O = (int)w.size(per_sample ? 1 : 0);
+  int I = (int)w.size(per_sample ? 2 : 1);
+  int kh = (int)w.size(per_sample ? 3 : 2);
+  int kw = (int)w.size(per_sample ? 4 : 3);
+  TORCH_CHECK(I == x.size(1), "conv2d: channel mismatch");
+  auto p = conv_params(x, O, kh, kw, (int)stride, (int)pad, per_sample);
+  auto out = torch::empty({p.B, O, p.OH, p.OW}, x.options());
+  if (x.scalar_type() == torch::kFloat32) {
+    gfa::launch_conv2d_fwd_f32(ptr<float>(out), cptr<float>(x),
+                               cptr<float>(w), p, cur_stream());
+  } else {
+    gfa::launch_conv2d_fwd_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
+                                cptr<gfa::bf16>(w), p, cur_stream());
+  }
+  return out;
+}
+
+Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
+                    int64_t kh, int64_t kw, bool per_sample) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 4);
+  TORCH_CHECK(x.scalar_type() == dy.scalar_type());
+  check_dtype(x, "conv2d_wgrad.x");
+  int O = (int)dy.size(1);
+  auto p = conv_params(x, O, (int)kh, (int)kw, (int)stride, (int)pad,
+                       per_sample);
+  TORCH_CHECK(p.OH == dy.size(2) && p.OW == dy.size(3),
+              "conv2d_wgrad: dy shape mismatch");
+  auto shape = per_sample
+                   ? std::vector<int64_t>{x.size(0), O, x.size(1), kh, kw}
+                   : std::vector<int64_t>{O, x.size(1), kh, kw};
+  auto dw = torch::empty(shape, x.options());
+  DISPATCH_FT(x, "conv2d_wgrad", {
+    gfa::launch_conv2d_wgrad<scalar_t>(ptr<scalar_t>(dw), cptr<scalar_t>(x),
+                                       cptr<scalar_t>(dy), p, cur_stream());
+  });
+  return dw;
+}
+
+Tensor mbstd(Tensor x, int64_t G, int64_t F, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  check_dtype(x, "mbstd.x");
+  int B = (int)x.size(0), C = (int)x.size(1);
+  int HW = (int)(x.size(2) * x.size(3));
+  TORCH_CHECK(B % G == 0 && C % F == 0 && G <= 32);
+  auto out = torch::empty({B / G, F},
+                          x.options().dtype(torch::kFloat32));
+  DISPATCH_FT(x, "mbstd", {
+    gfa::launch_mbstd<scalar_t>(out.data_ptr<float>(), cptr<scalar_t>(x), B,
+                                C, HW, (int)G, (int)F, (float)eps,
+                                cur_stream());
+  });
+  return out;
+}
+
+Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 3);
+  TORCH_CHECK(k.is_cuda() && k.is_contiguous() && k.dim() == 3);
+  TORCH_CHECK(v.is_cuda() && v.is_contiguous() && v.dim() == 3);
+  TORCH_CHECK(q.scalar_type() == k.scalar_type() &&
+              q.scalar_type() == v.scalar_type());
+  check_dtype(q, "bipartite_attn.q");
+  gfa::AttnParams p;
+  p.B = (int)q.size(0);
+  p.Nq = (int)q.size(1);
+  p.Nk = (int)k.size(1);
+  p.D = (int)q.size(2);
+  p.E = (int)v.size(2);
+  p.scale = (float)scale;
+  TORCH_CHECK(k.size(0) == p.B && v.size(0) == p.B);
+  TORCH_CHECK(k.size(2) == p.D && v.size(1) == p.Nk);
+  auto out = torch::empty({p.B, p.Nq, p.E}, q.options());
+  if (p.Nk <= 64) {
+    DISPATCH_FT(q, "attn", {
+      gfa::launch_attn_smalln<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(q),
+                                        cptr<scalar_t>(k), cptr<scalar_t>(v),
+                                        p, cur_stream());
+    });
+  } else {
+    TORCH_CHECK(p.Nq <= 64,
+                "bipartite_attn: long-N path requires Nq <= 64 "
+                "(bipartite attention is k x HW, never HW x HW)");
+    int nchunks = (p.Nk + 63) / 64;
+    auto f32opt = q.options().dtype(torch::kFloat32);
+    auto ws_m = torch::empty({p.B, nchunks, 64}, f32opt);
+    auto ws_l = torch::empty({p.B, nchunks, 64}, f32opt);
+    auto ws_o = torch::empty({(int64_t)p.B, nchunks, 64, p.E}, f32opt);
+    DISPATCH_FT(q, "attn", {
+      gfa::launch_attn_longn<scalar_t>(
+          ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k),
+          cptr<scalar_t>(v), ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
+          ws_o.data_ptr<float>(), p, nchunks, cur_stream());
+    });
+  }
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "gansformer_amd gfx950 HIP kernels";
+  m.def("fba", &fba, "fused bias+act (grad=0) / grad variant (grad=1)");
+  m.def("upfirdn2d", &upfirdn2d, "pad-upsample-FIR-downsample");
+  m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv2d (per-sample ok)");
+  m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight gradient");
+  m.def("mbstd", &mbstd, "minibatch stddev stats [B/G, F]");
+  m.def("bipartite_attn", &bipartite_attn, "softmax(QK^T)V");
+}

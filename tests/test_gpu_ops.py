@@ -1,0 +1,234 @@
+"""GPU kernel numerics: each HIP kernel vs the eager fp32 reference
+(SURVEY.md §4 item 1). fp32 tight tolerance, bf16 loose. Run via gpurun:
+    python -m pytest tests -m gpu -x -q
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    from gansformer_amd.ops import native
+    native.require_ext()  # fail loudly, never silently eager
+    return torch.device("cuda:0")
+
+
+def _C():
+    import gansformer_amd._C as C
+    return C
+
+
+def close(a, b, dtype):
+    a, b = a.float().cpu(), b.float().cpu()
+    tol = 1e-4 if dtype == torch.float32 else 5e-2
+    err = (a - b).abs().max().item()
+    scale = b.abs().max().item() + 1e-6
+    assert err / scale < tol, f"rel err {err / scale:.2e} (abs {err:.2e})"
+
+
+# ---------------- fba ----------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("act,code", [("linear", 0), ("lrelu", 2)])
+def test_fba_forward(dev, dtype, act, code):
+    from gansformer_amd.ops.fused_act import _eager_fwd
+    torch.manual_seed(0)
+    x = torch.randn(4, 16, 13, 13, device=dev, dtype=dtype)
+    b = torch.randn(16, device=dev, dtype=dtype)
+    y = _C().fba(x, b, torch.empty(0, device=dev, dtype=dtype), code, 0,
+                 0.2, math.sqrt(2.0), 256.0)
+    ref = _eager_fwd(x.float().cpu(), b.float().cpu(), act, 0.2,
+                     math.sqrt(2.0), 256.0)
+    close(y, ref, dtype)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fba_grad(dev, dtype):
+    from gansformer_amd.ops.fused_act import _eager_grad1
+    torch.manual_seed(1)
+    dy = torch.randn(2, 8, 9, 9, device=dev, dtype=dtype)
+    yref = torch.randn(2, 8, 9, 9, device=dev, dtype=dtype)
+    dx = _C().fba(dy, torch.empty(0, device=dev, dtype=dtype), yref, 2, 1,
+                  0.2, math.sqrt(2.0), 256.0)
+    ref = _eager_grad1(dy.float().cpu(), yref.float().cpu(), "lrelu", 0.2,
+                       math.sqrt(2.0), 256.0)
+    close(dx, ref, dtype)
+
+
+# ---------------- upfirdn2d ----------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("up,down,pad", [(1, 1, 2), (2, 1, 2), (1, 2, 1),
+                                         (2, 2, 3)])
+def test_upfirdn2d(dev, dtype, up, down, pad):
+    from gansformer_amd.ops.upfirdn2d import _eager_upfirdn2d, setup_filter
+    torch.manual_seed(2)
+    x = torch.randn(2, 5, 17, 23, device=dev, dtype=dtype)
+    f = setup_filter([1, 3, 3, 1], device=dev)
+    y = _C().upfirdn2d(x, f, up, up, down, down, pad, pad, pad, pad, 1.0)
+    ref = _eager_upfirdn2d(x.float().cpu(), f.cpu(), (up, up), (down, down),
+                           (pad, pad, pad, pad), 1.0)
+    assert y.shape == ref.shape
+    close(y, ref, dtype)
+
+
+def test_upfirdn2d_asym(dev):
+    from gansformer_amd.ops.upfirdn2d import _eager_upfirdn2d, setup_filter
+    x = torch.randn(1, 3, 11, 9, device=dev)
+    f = setup_filter([1, 2, 1], device=dev)
+    y = _C().upfirdn2d(x, f, 2, 1, 1, 2, 1, 2, 0, 1, 2.0)
+    ref = _eager_upfirdn2d(x.cpu(), f.cpu(), (1, 2), (2, 1), (0, 1, 1, 2), 2.0)
+    assert y.shape == ref.shape
+    close(y, ref, torch.float32)
+
+
+# ---------------- conv2d ----------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("per_sample", [False, True])
+@pytest.mark.parametrize("cfg", [
+    dict(I=16, O=32, H=16, k=3, stride=1, pad=1),
+    dict(I=8, O=8, H=9, k=3, stride=1, pad=1),       # odd sizes
+    dict(I=16, O=24, H=16, k=1, stride=1, pad=0),    # 1x1 (tRGB-like)
+    dict(I=3, O=32, H=16, k=1, stride=1, pad=0),     # fromRGB (K=3)
+    dict(I=16, O=32, H=16, k=3, stride=2, pad=1),    # downsample conv
+    dict(I=80, O=96, H=8, k=3, stride=1, pad=1),     # K not mult of 64
+])
+def test_conv2d_fwd(dev, dtype, per_sample, cfg):
+    torch.manual_seed(3)
+    B = 3
+    x = torch.randn(B, cfg["I"], cfg["H"], cfg["H"], device=dev, dtype=dtype)
+    wshape = (B, cfg["O"], cfg["I"], cfg["k"], cfg["k"]) if per_sample \
+        else (cfg["O"], cfg["I"], cfg["k"], cfg["k"])
+    w = torch.randn(*wshape, device=dev, dtype=dtype) * 0.1
+    y = _C().conv2d_fwd(x, w, cfg["stride"], cfg["pad"])
+    from gansformer_amd.ops.conv2d_grad import _eager_conv2d
+    ref = _eager_conv2d(x.float().cpu(), w.float().cpu(), cfg["stride"],
+                        cfg["pad"])
+    assert y.shape == ref.shape
+    close(y, ref, dtype)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("per_sample", [False, True])
+@pytest.mark.parametrize("stride", [1, 2])
+def test_conv2d_wgrad(dev, dtype, per_sample, stride):
+    torch.manual_seed(4)
+    B, I, O, H, k = 2, 12, 20, 12, 3
+    x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
+    OH = (H + 2 - k) // stride + 1
+    dy = torch.randn(B, O, OH, OH, device=dev, dtype=dtype)
+    dw = _C().conv2d_wgrad(x, dy, stride, 1, k, k, per_sample)
+    from gansformer_amd.ops.conv2d_grad import _eager_wgrad
+    ref = _eager_wgrad(x.float().cpu(), dy.float().cpu(), stride, 1, k, k,
+                       per_sample)
+    assert dw.shape == ref.shape
+    close(dw, ref, dtype)
+
+
+# ---------------- mbstd ----------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_mbstd(dev, dtype):
+    torch.manual_seed(5)
+    x = torch.randn(8, 16, 4, 4, device=dev, dtype=dtype)
+    stats = _C().mbstd(x, 4, 2, 1e-8)
+    assert stats.shape == (2, 2)
+    xf = x.float().cpu()
+    y = xf.reshape(4, 2, 2, 8, 4, 4)
+    y = y - y.mean(dim=0)
+    y = (y.square().mean(dim=0) + 1e-8).sqrt()
+    ref = y.mean(dim=[2, 3, 4])
+    close(stats, ref, dtype)
+
+
+# ---------------- bipartite attention ----------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("cfg", [
+    dict(B=2, Nq=100, Nk=16, D=64, E=48),     # simplex (small N)
+    dict(B=2, Nq=70, Nk=33, D=96, E=128),     # small N, odd sizes
+    dict(B=2, Nq=16, Nk=500, D=64, E=96),     # duplex reverse (long N)
+    dict(B=1, Nq=32, Nk=4096, D=128, E=256),  # long N, larger
+])
+def test_bipartite_attn(dev, dtype, cfg):
+    torch.manual_seed(6)
+    q = torch.randn(cfg["B"], cfg["Nq"], cfg["D"], device=dev, dtype=dtype)
+    k = torch.randn(cfg["B"], cfg["Nk"], cfg["D"], device=dev, dtype=dtype)
+    v = torch.randn(cfg["B"], cfg["Nk"], cfg["E"], device=dev, dtype=dtype)
+    scale = 1.0 / math.sqrt(cfg["D"])
+    out = _C().bipartite_attn(q, k, v, scale)
+    from gansformer_amd.ops.bipartite import _eager_attention
+    ref, _ = _eager_attention(q.float().cpu(), k.float().cpu(),
+                              v.float().cpu(), scale)
+    assert out.shape == ref.shape
+    close(out, ref, dtype)
+
+
+# ---------------- op-layer dispatch on GPU ----------------
+def test_ops_dispatch_native(dev):
+    """The Python op layer must route GPU tensors through the extension and
+    match the CPU eager path."""
+    from gansformer_amd.ops import bias_act, upfirdn2d, conv2d_gradfix, \
+        setup_filter, modulated_conv2d
+    torch.manual_seed(7)
+    x = torch.randn(2, 8, 16, 16)
+    b = torch.randn(8)
+    y_cpu = bias_act(x, b, act="lrelu")
+    y_gpu = bias_act(x.to(dev), b.to(dev), act="lrelu")
+    close(y_gpu, y_cpu, torch.float32)
+
+    f = setup_filter([1, 3, 3, 1])
+    y_cpu = upfirdn2d(x, f, up=2, padding=2)
+    y_gpu = upfirdn2d(x.to(dev), f.to(dev), up=2, padding=2)
+    close(y_gpu, y_cpu, torch.float32)
+
+    w = torch.randn(12, 8, 3, 3) * 0.2
+    s = torch.randn(2, 8)
+    y_cpu = modulated_conv2d(x, w, s)
+    y_gpu = modulated_conv2d(x.to(dev), w.to(dev), s.to(dev))
+    close(y_gpu, y_cpu, torch.float32)
+
+
+def test_conv_backward_gpu_vs_cpu(dev):
+    """Full autograd chain through the native kernels vs CPU eager."""
+    from gansformer_amd.ops import conv2d_gradfix
+    torch.manual_seed(8)
+    x0 = torch.randn(2, 8, 12, 12)
+    w0 = torch.randn(2, 12, 8, 3, 3) * 0.2
+
+    def run(devc):
+        x = x0.to(devc).requires_grad_(True)
+        w = w0.to(devc).requires_grad_(True)
+        y = conv2d_gradfix(x, w, stride=1, padding=1)
+        loss = (y.square()).sum()
+        gx, gw = torch.autograd.grad(loss, [x, w])
+        return gx.cpu(), gw.cpu()
+
+    gx_c, gw_c = run("cpu")
+    gx_g, gw_g = run(dev)
+    close(gx_g, gx_c, torch.float32)
+    close(gw_g, gw_c, torch.float32)
+
+
+def test_r1_double_backward_gpu(dev):
+    """R1 second-order replay through native fba/upfirdn/conv kernels."""
+    from gansformer_amd.models.networks import Discriminator
+    from gansformer_amd.training.loss import r1_penalty
+    torch.manual_seed(9)
+    D = Discriminator(img_resolution=32, channel_base=1024, channel_max=64,
+                      mbstd_group_size=2, bf16_res_count=0).to(dev)
+    real = torch.randn(2, 3, 32, 32, device=dev, requires_grad=True)
+    r1 = r1_penalty(D(real), real)
+    r1.backward()
+    grads = [p.grad for p in D.parameters() if p.grad is not None]
+    assert grads and all(torch.isfinite(g).all() for g in grads)
+
+    # compare against CPU
+    Dc = Discriminator(img_resolution=32, channel_base=1024, channel_max=64,
+                       mbstd_group_size=2, bf16_res_count=0)
+    Dc.load_state_dict({k: v.cpu() for k, v in D.state_dict().items()})
+    realc = real.detach().cpu().requires_grad_(True)
+    r1c = r1_penalty(Dc(realc), realc)
+    assert abs(r1.item() - r1c.item()) / (abs(r1c.item()) + 1e-6) < 1e-3
