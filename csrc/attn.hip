@@ -15,56 +15,7 @@
 
 namespace gfa {
 
-// ---- per-dtype LDS tile ops: [row][BK] with 8-elem XOR swizzle (bf16)
-// or +1 padding (f32); fragment read helpers for the two MFMA families.
-template <typename T> struct TileOps;
-
-template <> struct TileOps<bf16> {
-  static constexpr int BK = 32;          // K depth per tile
-  static constexpr int ROW = 32;         // u16 elements per LDS row
-  using elem = u16;
-  GFA_DEV static int idx(int row, int k) {
-    int g = k >> 3;
-    return row * ROW + ((g ^ (row & 3)) << 3) + (k & 7);
-  }
-  GFA_DEV static void store(elem* lds, int row, int k, float v) {
-    lds[idx(row, k)] = f32_to_bf16_bits(v);
-  }
-  // one 16x16x32 MFMA: A rows arow.. (16), B rows brow.. (16), full BK
-  GFA_DEV static f32x4 mfma(const elem* As, const elem* Bs, int arow0,
-                            int brow0, int lane, f32x4 acc) {
-    int ar = arow0 + (lane & 15);
-    int br = brow0 + (lane & 15);
-    int g = lane >> 4;
-    s16x8 af = *reinterpret_cast<const s16x8*>(
-        &As[ar * ROW + ((g ^ (ar & 3)) << 3)]);
-    s16x8 bf = *reinterpret_cast<const s16x8*>(
-        &Bs[br * ROW + ((g ^ (br & 3)) << 3)]);
-    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
-  }
-};
-
-template <> struct TileOps<float> {
-  static constexpr int BK = 32;
-  static constexpr int ROW = 33;         // +1 pad breaks bank conflicts
-  using elem = float;
-  GFA_DEV static int idx(int row, int k) { return row * ROW + k; }
-  GFA_DEV static void store(elem* lds, int row, int k, float v) {
-    lds[idx(row, k)] = v;
-  }
-  GFA_DEV static f32x4 mfma(const elem* As, const elem* Bs, int arow0,
-                            int brow0, int lane, f32x4 acc) {
-    int ar = arow0 + (lane & 15);
-    int br = brow0 + (lane & 15);
-#pragma unroll
-    for (int kk = 0; kk < BK; kk += 4) {
-      float av = As[ar * ROW + kk + (lane >> 4)];
-      float bv = Bs[br * ROW + kk + (lane >> 4)];
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc, 0, 0, 0);
-    }
-    return acc;
-  }
-};
+// per-dtype LDS tile ops (TileOps) are shared in common.h.
 
 struct AttnParams {
   int B, Nq, Nk, D, E;

@@ -219,20 +219,25 @@ void conv2d_fwd_f32(float* __restrict__ y, const float* __restrict__ x,
 
 // ---------------- wgrad ----------------
 // dW[b?][o][kw_idx] = sum_p dY[b][o][p] * patch[b][kw_idx][p]
-// A rows: dY (contiguous in p). B rows: implicit patches (contiguous-ish).
-// Shared weights: K loop covers (b, p); per-sample: fixed b per block.
+// GEMM: M = O (dY rows, p-contiguous -> 16B vector staging), N = Kw
+// (implicit patches, 16B staging when the 8-pixel run is contiguous and
+// in-bounds), K = P (x B for shared weights). MFMA via TileOps: bf16 path
+// on v_mfma_f32_16x16x32_bf16, f32 on the exact-f32 16x16x4.
 template <typename T>
 __global__ __launch_bounds__(256)
 void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
                          const T* __restrict__ dy, ConvParams p) {
-  constexpr int BM = 64, BN = 64, BK = 32;
-  __shared__ float As[BM][BK / 8][9];   // [row][granule][8+1 pad] as floats
-  __shared__ float Bs[BN][BK / 8][9];
+  using TO = TileOps<T>;
+  constexpr int BK = TO::BK;   // pixels per K step (32)
+  constexpr int ROW = TO::ROW;
+  using elem = typename TO::elem;
+  __shared__ elem As[64 * ROW];
+  __shared__ elem Bs[64 * ROW];
 
   const int Kw = p.I * p.kh * p.kw;
   const int P = p.OH * p.OW;
-  const int m0 = blockIdx.y * BM;      // O tile
-  const int n0 = blockIdx.x * BN;      // Kw tile
+  const int m0 = blockIdx.y * 64;      // O tile
+  const int n0 = blockIdx.x * 64;      // Kw tile
   const int nb = p.per_sample ? 1 : p.B;
   const int b_fix = p.per_sample ? blockIdx.z : 0;
 
@@ -242,6 +247,14 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
   const int wm = (wave >> 1) * 32;
   const int wn = (wave & 1) * 32;
 
+  // per-thread staging roles (constant over the K loop)
+  const int srow = t >> 2;             // LDS row this thread fills
+  const int sg = t & 3;                // granule (8 K-elements)
+  const int b_kwi = n0 + srow;
+  const int b_i = b_kwi / (p.kh * p.kw);
+  const int b_rs = b_kwi % (p.kh * p.kw);
+  const int b_r = b_rs / p.kw, b_s = b_rs % p.kw;
+
   f32x4 acc[2][2] = {};
 
   for (int bb = 0; bb < nb; ++bb) {
@@ -249,70 +262,94 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
     const T* dyb = dy + (long)b * p.O * P;
     const T* xb = x + (long)b * p.I * p.H * p.W;
     for (int p0 = 0; p0 < P; p0 += BK) {
-      {  // stage A (dY): row t/4, granule t%4 -> 8 p values
-        int row = t >> 2, g = t & 3;
-        int o = m0 + row;
+      {  // ---- stage A (dY): always p-contiguous in memory
+        int o = m0 + srow;
+        int pp = p0 + sg * 8;
+        if constexpr (sizeof(T) == 2) {
+          if (o < p.O && pp + 8 <= P && (((long)o * P + pp) & 7) == 0) {
+            TO::store_vec8(As, srow, sg,
+                           *reinterpret_cast<const s16x8*>(
+                               reinterpret_cast<const u16*>(dyb) +
+                               (long)o * P + pp));
+          } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int pp = p0 + g * 8 + j;
-          As[row][g][j] = (o < p.O && pp < P)
-                              ? to_f32(dyb[(long)o * P + pp]) : 0.f;
+            for (int j = 0; j < 8; ++j)
+              TO::store(As, srow, sg * 8 + j,
+                        (o < p.O && pp + j < P)
+                            ? to_f32(dyb[(long)o * P + pp + j]) : 0.f);
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            TO::store(As, srow, sg * 8 + j,
+                      (o < p.O && pp + j < P)
+                          ? to_f32(dyb[(long)o * P + pp + j]) : 0.f);
         }
       }
-      {  // stage B (patches): row = Kw index, same granule layout
-        int row = t >> 2, g = t & 3;
-        int kwi = n0 + row;
-        int i = kwi / (p.kh * p.kw);
-        int rs = kwi % (p.kh * p.kw);
-        int r = rs / p.kw, s = rs % p.kw;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int pp = p0 + g * 8 + j;
-          float v = 0.f;
-          if (kwi < Kw && pp < P) {
-            int oy = pp / p.OW, ox = pp % p.OW;
-            int iy = oy * p.stride + r - p.pad;
-            int ix = ox * p.stride + s - p.pad;
-            if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
-              v = to_f32(xb[((long)i * p.H + iy) * p.W + ix]);
+      {  // ---- stage B (patches): vector when the run is one x row
+        int pp = p0 + sg * 8;
+        int oy0 = pp / p.OW, ox0 = pp % p.OW;
+        bool fast = false;
+        long src = 0;
+        if (b_kwi < Kw && pp + 7 < P && p.stride == 1 &&
+            (pp + 7) / p.OW == oy0) {
+          int iy = oy0 + b_r - p.pad;
+          int ix = ox0 + b_s - p.pad;
+          if (iy >= 0 && iy < p.H && ix >= 0 && ix + 7 < p.W) {
+            fast = true;
+            src = ((long)b_i * p.H + iy) * p.W + ix;
           }
-          Bs[row][g][j] = v;
+        }
+        if constexpr (sizeof(T) == 2) {
+          if (fast && (src & 7) == 0) {
+            TO::store_vec8(Bs, srow, sg,
+                           *reinterpret_cast<const s16x8*>(
+                               reinterpret_cast<const u16*>(xb) + src));
+            fast = true;
+          } else {
+            fast = false;
+          }
+        } else {
+          fast = false;
+        }
+        if (!fast) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int pj = pp + j;
+            float v = 0.f;
+            if (b_kwi < Kw && pj < P) {
+              int oy = pj / p.OW, ox = pj % p.OW;
+              int iy = oy * p.stride + b_r - p.pad;
+              int ix = ox * p.stride + b_s - p.pad;
+              if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
+                v = to_f32(xb[((long)b_i * p.H + iy) * p.W + ix]);
+            }
+            TO::store(Bs, srow, sg * 8 + j, v);
+          }
         }
       }
       __syncthreads();
-      // f32 MFMA over the staged K-slab (values may be bf16-origin; the
-      // products are exact in f32)
 #pragma unroll
-      for (int g = 0; g < 4; ++g) {
+      for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int kk = 0; kk < 8; kk += 4) {
-#pragma unroll
-          for (int mi = 0; mi < 2; ++mi)
-#pragma unroll
-            for (int ni = 0; ni < 2; ++ni) {
-              float av = As[wm + mi * 16 + (lane & 15)][g][kk + (lane >> 4)];
-              float bv = Bs[wn + ni * 16 + (lane & 15)][g][kk + (lane >> 4)];
-              acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                  av, bv, acc[mi][ni], 0, 0, 0);
-            }
-        }
-      }
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = TO::mfma(As, Bs, wm + mi * 16, wn + ni * 16, lane,
+                                 acc[mi][ni]);
       __syncthreads();
     }
   }
-  const int Kw2 = Kw;
-  T* dwb = dw + (p.per_sample ? (long)b_fix * p.O * Kw2 : 0);
+  T* dwb = dw + (p.per_sample ? (long)b_fix * p.O * Kw : 0);
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       int col = n0 + wn + ni * 16 + (lane & 15);
-      if (col >= Kw2) continue;
+      if (col >= Kw) continue;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
         if (row < p.O)
-          dwb[(long)row * Kw2 + col] = from_f32<T>(acc[mi][ni][reg]);
+          dwb[(long)row * Kw + col] = from_f32<T>(acc[mi][ni][reg]);
       }
     }
 }

@@ -379,7 +379,7 @@ class SynthesisNetwork(nn.Module):
     def __init__(self, w_dim=512, img_resolution=256, img_channels=3,
                  num_components=16, transformer="duplex", integration="mul",
                  attn_resolutions=None, use_pos=True, channel_base=32768,
-                 channel_max=512, bf16_res_count=4, conv_clamp=256.0):
+                 channel_max=512, bf16_res_count=16, conv_clamp=256.0):
         super().__init__()
         self.w_dim = w_dim
         self.img_resolution = img_resolution
@@ -433,7 +433,7 @@ class Generator(nn.Module):
     def __init__(self, z_dim=512, w_dim=512, img_resolution=256, img_channels=3,
                  num_components=16, transformer="duplex", integration="mul",
                  attn_resolutions=None, use_pos=True, channel_base=32768,
-                 channel_max=512, bf16_res_count=4, mapping_layers=8,
+                 channel_max=512, bf16_res_count=16, mapping_layers=8,
                  mapping_lr_mul=0.01, w_avg_beta=0.995, conv_clamp=256.0,
                  style_mixing_prob=0.9):
         super().__init__()
@@ -516,7 +516,7 @@ class DiscriminatorBlock(nn.Module):
 class Discriminator(nn.Module):
     def __init__(self, img_resolution=256, img_channels=3, channel_base=32768,
                  channel_max=512, mbstd_group_size=4, mbstd_num_channels=1,
-                 bf16_res_count=4, conv_clamp=256.0):
+                 bf16_res_count=16, conv_clamp=256.0):
         super().__init__()
         self.img_resolution = img_resolution
         self.img_channels = img_channels

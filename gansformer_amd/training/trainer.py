@@ -19,13 +19,14 @@ class GANTrainer:
                  g_lr=0.002, d_lr=0.002, beta1=0.0, beta2=0.99, eps=1e-8,
                  gamma=10.0, d_reg_interval=16, g_reg_interval=8,
                  pl_weight=2.0, pl_decay=0.01, ema_kimg=10.0,
-                 ema_rampup=0.05, rounds=1, bucket_mb=25.0):
+                 ema_rampup=0.05, rounds=1, bucket_mb=25.0, r1_fp32=False):
         self.G, self.D, self.Gs = G, D, Gs
         self.device = device
         self.batch_gpu = batch_gpu
         self.batch_size = batch_size
         self.rounds = rounds
         self.gamma = gamma
+        self.r1_fp32 = r1_fp32
         self.d_reg_interval = d_reg_interval
         self.g_reg_interval = g_reg_interval
         self.ema_kimg = ema_kimg
@@ -71,7 +72,7 @@ class GANTrainer:
             self.d_red.prepare()
             for _ in range(rounds):
                 real = next_real_fn().requires_grad_(True)
-                real_logits = D(real, force_fp32=True)
+                real_logits = D(real, force_fp32=self.r1_fp32)
                 r1 = r1_penalty(real_logits, real)
                 ((self.gamma * r1 * self.d_reg_interval) / rounds).backward()
                 rep("Loss/r1", r1.item())

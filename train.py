@@ -90,7 +90,7 @@ def main(argv=None):
     p.add_argument("--fmap-max", type=int, default=512)
     p.add_argument("--mapping-layers", type=int, default=8)
     p.add_argument("--mbstd-group", type=int, default=4)
-    p.add_argument("--bf16-res", type=int, default=4,
+    p.add_argument("--bf16-res", type=int, default=16,
                    help="number of top resolutions computed in bf16")
     # loss / sched
     p.add_argument("--gamma", type=float, default=10.0, help="R1 weight")
