@@ -224,8 +224,9 @@ void conv2d_fwd_f32(float* __restrict__ y, const float* __restrict__ x,
 // on v_mfma_f32_16x16x32_bf16, f32 on the exact-f32 16x16x4.
 template <typename T>
 __global__ __launch_bounds__(256)
-void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
-                         const T* __restrict__ dy, ConvParams p) {
+void conv2d_wgrad_kernel(T* __restrict__ dw, float* __restrict__ ws,
+                         const T* __restrict__ x, const T* __restrict__ dy,
+                         ConvParams p, int nsplit) {
   using TO = TileOps<T>;
   constexpr int BK = TO::BK;   // pixels per K step (32)
   constexpr int ROW = TO::ROW;
@@ -238,7 +239,14 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
   const int m0 = blockIdx.y * 64;      // O tile
   const int n0 = blockIdx.x * 64;      // Kw tile
   const int nb = p.per_sample ? 1 : p.B;
-  const int b_fix = p.per_sample ? blockIdx.z : 0;
+  const int b_fix = p.per_sample ? (int)blockIdx.z / nsplit : 0;
+  // split-K: this block covers K-iterations [it0, it1) of nb * ceil(P/BK)
+  const int split = p.per_sample ? (int)blockIdx.z % nsplit : (int)blockIdx.z;
+  const int iters_per_b = (P + BK - 1) / BK;
+  const long total_iters = (long)nb * iters_per_b;
+  const long span = (total_iters + nsplit - 1) / nsplit;
+  const long it0 = split * span;
+  const long it1 = min(total_iters, it0 + span);
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -256,11 +264,13 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
 
   f32x4 acc[2][2] = {};
 
-  for (int bb = 0; bb < nb; ++bb) {
+  for (long it = it0; it < it1; ++it) {
+    const int bb = (int)(it / iters_per_b);
+    const int p0 = (int)(it % iters_per_b) * BK;
     const int b = p.per_sample ? b_fix : bb;
     const T* dyb = dy + (long)b * p.O * P;
     const T* xb = x + (long)b * p.I * p.H * p.W;
-    for (int p0 = 0; p0 < P; p0 += BK) {
+    {
       {  // ---- stage A (dY): always p-contiguous in memory
         int o = m0 + srow;
         int pp = p0 + sg * 8;
@@ -337,7 +347,7 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
       __syncthreads();
     }
   }
-  T* dwb = dw + (p.per_sample ? (long)b_fix * p.O * Kw : 0);
+  const long wbase = p.per_sample ? (long)b_fix * p.O * Kw : 0;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -347,8 +357,12 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, const T* __restrict__ x,
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
-        if (row < p.O)
-          dwb[(long)row * Kw + col] = from_f32<T>(acc[mi][ni][reg]);
+        if (row < p.O) {
+          if (nsplit == 1)
+            dw[wbase + (long)row * Kw + col] = from_f32<T>(acc[mi][ni][reg]);
+          else
+            atomicAdd(&ws[wbase + (long)row * Kw + col], acc[mi][ni][reg]);
+        }
       }
     }
 }
@@ -364,17 +378,33 @@ void launch_conv2d_fwd_f32(float* y, const float* x, const float* w,
   dim3 grid(ceil_div(p.OH * (long)p.OW, 64), ceil_div(p.O, 64), p.B);
   hipLaunchKernelGGL(conv2d_fwd_f32, grid, dim3(256), 0, s, y, x, w, p);
 }
-template <typename T>
-void launch_conv2d_wgrad(T* dw, const T* x, const T* dy, const ConvParams& p,
-                         hipStream_t s) {
+int conv2d_wgrad_nsplit(const ConvParams& p) {
+  // choose split-K so the grid has >~1024 workgroups (256 CUs x 4)
   int Kw = p.I * p.kh * p.kw;
-  dim3 grid(ceil_div(Kw, 64), ceil_div(p.O, 64), p.per_sample ? p.B : 1);
-  hipLaunchKernelGGL(conv2d_wgrad_kernel<T>, grid, dim3(256), 0, s, dw, x,
-                     dy, p);
+  long natural = (long)ceil_div(Kw, 64) * ceil_div(p.O, 64)
+                 * (p.per_sample ? p.B : 1);
+  int iters = ceil_div((long)p.OH * p.OW, 32) * (p.per_sample ? 1 : p.B);
+  int nsplit = (int)((1024 + natural - 1) / natural);
+  if (nsplit > iters) nsplit = iters;
+  if (nsplit > 64) nsplit = 64;
+  if (nsplit < 1) nsplit = 1;
+  return nsplit;
 }
-template void launch_conv2d_wgrad<float>(float*, const float*, const float*,
-                                         const ConvParams&, hipStream_t);
-template void launch_conv2d_wgrad<bf16>(bf16*, const bf16*, const bf16*,
-                                        const ConvParams&, hipStream_t);
+
+template <typename T>
+void launch_conv2d_wgrad(T* dw, float* ws, const T* x, const T* dy,
+                         const ConvParams& p, int nsplit, hipStream_t s) {
+  int Kw = p.I * p.kh * p.kw;
+  dim3 grid(ceil_div(Kw, 64), ceil_div(p.O, 64),
+            (p.per_sample ? p.B : 1) * nsplit);
+  hipLaunchKernelGGL(conv2d_wgrad_kernel<T>, grid, dim3(256), 0, s, dw, ws,
+                     x, dy, p, nsplit);
+}
+template void launch_conv2d_wgrad<float>(float*, float*, const float*,
+                                         const float*, const ConvParams&,
+                                         int, hipStream_t);
+template void launch_conv2d_wgrad<bf16>(bf16*, float*, const bf16*,
+                                        const bf16*, const ConvParams&, int,
+                                        hipStream_t);
 
 }  // namespace gfa

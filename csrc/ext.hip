@@ -41,9 +41,10 @@ void launch_conv2d_fwd_bf16(bf16*, const bf16*, const bf16*,
                             const ConvParams&, hipStream_t);
 void launch_conv2d_fwd_f32(float*, const float*, const float*,
                            const ConvParams&, hipStream_t);
+int conv2d_wgrad_nsplit(const ConvParams&);
 template <typename T>
-void launch_conv2d_wgrad(T*, const T*, const T*, const ConvParams&,
-                         hipStream_t);
+void launch_conv2d_wgrad(T*, float*, const T*, const T*, const ConvParams&,
+                         int, hipStream_t);
 template <typename T>
 void launch_mbstd(float*, const T*, int, int, int, int, int, float,
                   hipStream_t);
@@ -195,12 +196,23 @@ Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
   auto shape = per_sample
                    ? std::vector<int64_t>{x.size(0), O, x.size(1), kh, kw}
                    : std::vector<int64_t>{O, x.size(1), kh, kw};
-  auto dw = torch::empty(shape, x.options());
+  int nsplit = gfa::conv2d_wgrad_nsplit(p);
+  if (nsplit == 1) {
+    auto dw = torch::empty(shape, x.options());
+    DISPATCH_FT(x, "conv2d_wgrad", {
+      gfa::launch_conv2d_wgrad<scalar_t>(ptr<scalar_t>(dw), nullptr,
+                                         cptr<scalar_t>(x), cptr<scalar_t>(dy),
+                                         p, 1, cur_stream());
+    });
+    return dw;
+  }
+  auto ws = torch::zeros(shape, x.options().dtype(torch::kFloat32));
   DISPATCH_FT(x, "conv2d_wgrad", {
-    gfa::launch_conv2d_wgrad<scalar_t>(ptr<scalar_t>(dw), cptr<scalar_t>(x),
-                                       cptr<scalar_t>(dy), p, cur_stream());
+    gfa::launch_conv2d_wgrad<scalar_t>(nullptr, ws.data_ptr<float>(),
+                                       cptr<scalar_t>(x), cptr<scalar_t>(dy),
+                                       p, nsplit, cur_stream());
   });
-  return dw;
+  return ws.to(x.scalar_type());
 }
 
 Tensor mbstd(Tensor x, int64_t G, int64_t F, double eps) {

@@ -550,7 +550,9 @@ class Discriminator(nn.Module):
         self.out = FullyConnected(prev_ch, 1)
 
     def forward(self, img, force_fp32=False):
-        x = self.frgb(img.to(torch.float32))
+        blk0_bf16 = (self.blocks and self.blocks[0].use_bf16
+                     and not force_fp32 and img.is_cuda)
+        x = self.frgb(img.to(torch.bfloat16 if blk0_bf16 else torch.float32))
         for blk in self.blocks:
             x = blk(x, force_fp32=force_fp32)
         x = x.to(torch.float32)
