@@ -135,6 +135,111 @@ void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
     }
 }
 
+// ---------------- bf16 forward, 128x128 tile ----------------
+// For large pixel counts: 4 waves, each owning a 64x64 sub-tile (4x4
+// fragments), BK=32. 16 MFMA per 8 fragment ds_read_b128 per wave per
+// K-step — 4x the MFMA:staging ratio of the 64x64 kernel.
+__global__ __launch_bounds__(256)
+void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
+                         const bf16* __restrict__ w, ConvParams p) {
+  constexpr int BM = 128, BN = 128, BK = 32;
+  __shared__ u16 As[BM * BK];
+  __shared__ u16 Bs[BN * BK];
+
+  const int K = p.I * p.kh * p.kw;
+  const int P = p.OH * p.OW;
+  const int b = blockIdx.z;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  const bf16* wb = w + (p.per_sample ? (long)b * p.O * K : 0);
+  const bf16* xb = x + (long)b * p.I * p.H * p.W;
+  bf16* yb = y + (long)b * p.O * P;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int khw = p.kh * p.kw;
+
+  f32x4 acc[4][4] = {};
+  const bool a_vec = (K % 8 == 0);
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A: 128 rows x 32 k; thread t covers rows t/4 and 64+t/4
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int row = half * 64 + (t >> 2), g = t & 3;
+      int o = m0 + row;
+      int kk = k0 + g * 8;
+      u16* dst = &As[row * BK + (swz(row, g) << 3)];
+      if (o < p.O && kk + 8 <= K && a_vec) {
+        *reinterpret_cast<s16x8*>(dst) =
+            *reinterpret_cast<const s16x8*>(wb + (long)o * K + kk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16 v = 0;
+          if (o < p.O && kk + j < K)
+            v = __builtin_bit_cast(u16, wb[(long)o * K + kk + j]);
+          dst[j] = v;
+        }
+      }
+    }
+    // ---- stage B: 128 pixels x 32 k; idx = t + 256*j -> coalesced in pix
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int idx = t + 256 * j;
+      int lk = idx >> 7;         // k within tile
+      int lp = idx & 127;        // pixel within tile
+      int kk = k0 + lk;
+      int pix = n0 + lp;
+      u16 v = 0;
+      if (kk < K && pix < P) {
+        int i = kk / khw;
+        int rs = kk % khw;
+        int r = rs / p.kw, s = rs % p.kw;
+        int oy = pix / p.OW, ox = pix % p.OW;
+        int iy = oy * p.stride + r - p.pad;
+        int ix = ox * p.stride + s - p.pad;
+        if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
+          v = __builtin_bit_cast(u16, xb[((long)i * p.H + iy) * p.W + ix]);
+      }
+      Bs[lp * BK + (swz(lp, lk >> 3) << 3) + (lk & 7)] = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      int arow = wm + mi * 16 + (lane & 15);
+      int ag = lane >> 4;
+      s16x8 af = *reinterpret_cast<const s16x8*>(
+          &As[arow * BK + (swz(arow, ag) << 3)]);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int brow = wn + ni * 16 + (lane & 15);
+        s16x8 bfr = *reinterpret_cast<const s16x8*>(
+            &Bs[brow * BK + (swz(brow, ag) << 3)]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bfr, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = n0 + wn + ni * 16 + (lane & 15);
+      if (col >= P) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        if (row < p.O)
+          yb[(long)row * P + col] = from_f32<bf16>(acc[mi][ni][reg]);
+      }
+    }
+}
+
 // ---------------- f32 forward (exact f32, 16x16x4 MFMA) ----------------
 __global__ __launch_bounds__(256)
 void conv2d_fwd_f32(float* __restrict__ y, const float* __restrict__ x,
@@ -371,8 +476,15 @@ void conv2d_wgrad_kernel(T* __restrict__ dw, float* __restrict__ ws,
 // ---------------- launchers ----------------
 void launch_conv2d_fwd_bf16(bf16* y, const bf16* x, const bf16* w,
                             const ConvParams& p, hipStream_t s) {
-  dim3 grid(ceil_div(p.OH * (long)p.OW, 64), ceil_div(p.O, 64), p.B);
-  hipLaunchKernelGGL(conv2d_fwd_bf16, grid, dim3(256), 0, s, y, x, w, p);
+  long P = (long)p.OH * p.OW;
+  if (P >= 4096) {
+    dim3 grid(ceil_div(P, 128), ceil_div(p.O, 128), p.B);
+    hipLaunchKernelGGL(conv2d_fwd_bf16_128, grid, dim3(256), 0, s, y, x, w,
+                       p);
+  } else {
+    dim3 grid(ceil_div(P, 64), ceil_div(p.O, 64), p.B);
+    hipLaunchKernelGGL(conv2d_fwd_bf16, grid, dim3(256), 0, s, y, x, w, p);
+  }
 }
 void launch_conv2d_fwd_f32(float* y, const float* x, const float* w,
                            const ConvParams& p, hipStream_t s) {
