@@ -161,6 +161,15 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
   const int wn = (wave & 1) * 64;
   const int khw = p.kh * p.kw;
 
+  // precompute this thread's two staged pixels' base coords (fixed for
+  // the whole K loop — kills the per-element div/mod that VALU-bound v1)
+  int pix_a = n0 + lane, pix_b = n0 + 64 + lane;
+  int iya = (pix_a / p.OW) * p.stride - p.pad;
+  int ixa = (pix_a % p.OW) * p.stride - p.pad;
+  int iyb = (pix_b / p.OW) * p.stride - p.pad;
+  int ixb = (pix_b % p.OW) * p.stride - p.pad;
+  const bool va = pix_a < P, vb = pix_b < P;
+
   f32x4 acc[4][4] = {};
   const bool a_vec = (K % 8 == 0);
 
@@ -185,26 +194,27 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
         }
       }
     }
-    // ---- stage B: 128 pixels x 32 k; idx = t + 256*j -> coalesced in pix
+    // ---- stage B: 128 pixels x 32 k; wave w owns k range [w*8, w*8+8),
+    // lanes sweep pixels (coalesced 64-wide); tap decode is wave-uniform.
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      int idx = t + 256 * j;
-      int lk = idx >> 7;         // k within tile
-      int lp = idx & 127;        // pixel within tile
+    for (int kj = 0; kj < 8; ++kj) {
+      int lk = wave * 8 + kj;
       int kk = k0 + lk;
-      int pix = n0 + lp;
-      u16 v = 0;
-      if (kk < K && pix < P) {
-        int i = kk / khw;
+      u16 va16 = 0, vb16 = 0;
+      if (kk < K) {
+        int i = kk / khw;              // wave-uniform (SALU)
         int rs = kk % khw;
         int r = rs / p.kw, s = rs % p.kw;
-        int oy = pix / p.OW, ox = pix % p.OW;
-        int iy = oy * p.stride + r - p.pad;
-        int ix = ox * p.stride + s - p.pad;
-        if (iy >= 0 && iy < p.H && ix >= 0 && ix < p.W)
-          v = __builtin_bit_cast(u16, xb[((long)i * p.H + iy) * p.W + ix]);
+        const bf16* xi = xb + (long)i * p.H * p.W;
+        int ya = iya + r, xa = ixa + s;
+        if (va && ya >= 0 && ya < p.H && xa >= 0 && xa < p.W)
+          va16 = __builtin_bit_cast(u16, xi[(long)ya * p.W + xa]);
+        int yb2 = iyb + r, xb2 = ixb + s;
+        if (vb && yb2 >= 0 && yb2 < p.H && xb2 >= 0 && xb2 < p.W)
+          vb16 = __builtin_bit_cast(u16, xi[(long)yb2 * p.W + xb2]);
       }
-      Bs[lp * BK + (swz(lp, lk >> 3) << 3) + (lk & 7)] = v;
+      Bs[lane * BK + (swz(lane, lk >> 3) << 3) + (lk & 7)] = va16;
+      Bs[(64 + lane) * BK + (swz(64 + lane, lk >> 3) << 3) + (lk & 7)] = vb16;
     }
     __syncthreads();
 #pragma unroll
