@@ -60,29 +60,28 @@ template <> GFA_DEV u16 f32_to_raw<u16>(float v) {
 template <typename T> struct TileOps;
 
 template <> struct TileOps<bf16> {
+  // Row stride 40 u16 = 80 B = 20 dwords: rows 0..15 start at 16 distinct
+  // bank quads ((row*20) % 64 is a bijection onto {0,4,...,60}), so the
+  // 16-lane ds_read_b128 fragment reads are conflict-free, and every row
+  // base stays 16-B aligned (Guideline 17). Scalar b16 writes are <=2-way.
   static constexpr int BK = 32;   // K depth per tile
-  static constexpr int ROW = 32;  // u16 elements per LDS row
+  static constexpr int ROW = 40;  // u16 elements per LDS row (32 + 8 pad)
   using elem = u16;
-  GFA_DEV static int idx(int row, int k) {
-    int g = k >> 3;
-    return row * ROW + ((g ^ (row & 3)) << 3) + (k & 7);
-  }
+  GFA_DEV static int idx(int row, int k) { return row * ROW + k; }
   GFA_DEV static void store(elem* lds, int row, int k, float v) {
-    lds[idx(row, k)] = f32_to_bf16_bits(v);
+    lds[row * ROW + k] = f32_to_bf16_bits(v);
   }
-  // store a full 8-element granule (k8 = granule index in [0, 4))
+  // store a full 8-element granule (g = granule index in [0, 4))
   GFA_DEV static void store_vec8(elem* lds, int row, int g, s16x8 v) {
-    *reinterpret_cast<s16x8*>(&lds[row * ROW + ((g ^ (row & 3)) << 3)]) = v;
+    *reinterpret_cast<s16x8*>(&lds[row * ROW + (g << 3)]) = v;
   }
   GFA_DEV static f32x4 mfma(const elem* As, const elem* Bs, int arow0,
                             int brow0, int lane, f32x4 acc) {
     int ar = arow0 + (lane & 15);
     int br = brow0 + (lane & 15);
     int g = lane >> 4;
-    s16x8 af = *reinterpret_cast<const s16x8*>(
-        &As[ar * ROW + ((g ^ (ar & 3)) << 3)]);
-    s16x8 bfr = *reinterpret_cast<const s16x8*>(
-        &Bs[br * ROW + ((g ^ (br & 3)) << 3)]);
+    s16x8 af = *reinterpret_cast<const s16x8*>(&As[ar * ROW + (g << 3)]);
+    s16x8 bfr = *reinterpret_cast<const s16x8*>(&Bs[br * ROW + (g << 3)]);
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc, 0, 0, 0);
   }
 };

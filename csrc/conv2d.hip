@@ -37,9 +37,9 @@ __device__ __forceinline__ int swz(int row, int g) { return g ^ (row & 3); }
 __global__ __launch_bounds__(256)
 void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
                      const bf16* __restrict__ w, ConvParams p) {
-  constexpr int BM = 64, BN = 64, BK = 32;
-  __shared__ u16 As[BM * BK];
-  __shared__ u16 Bs[BN * BK];
+  constexpr int BM = 64, BN = 64, BK = 32, RP = 40;  // padded row (bank-safe)
+  __shared__ u16 As[BM * RP];
+  __shared__ u16 Bs[BN * RP];
 
   const int K = p.I * p.kh * p.kw;
   const int P = p.OH * p.OW;
@@ -65,7 +65,7 @@ void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
       int row = t >> 2, g = t & 3;
       int o = m0 + row;
       int kk = k0 + g * 8;
-      u16* dst = &As[row * BK + swz(row, g) * 8];
+      u16* dst = &As[row * RP + g * 8];
       if (o < p.O && kk + 8 <= K && a_vec) {
         *reinterpret_cast<s16x8*>(dst) =
             *reinterpret_cast<const s16x8*>(wb + (long)o * K + kk);
@@ -97,7 +97,7 @@ void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
       }
       int lk = kk - k0;
       int lp = pix - n0;
-      Bs[lp * BK + (swz(lp, lk >> 3) << 3) + (lk & 7)] = v;
+      Bs[lp * RP + lk] = v;
     }
     __syncthreads();
     // ---- MFMA: each wave 2x2 fragments of 16x16, K=32 in one mfma each
@@ -108,10 +108,10 @@ void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
         int arow = wm + mi * 16 + (lane & 15);
         int ag = lane >> 4;
         s16x8 af = *reinterpret_cast<const s16x8*>(
-            &As[arow * BK + swz(arow, ag) * 8]);
+            &As[arow * RP + ag * 8]);
         int brow = wn + ni * 16 + (lane & 15);
         s16x8 bf = *reinterpret_cast<const s16x8*>(
-            &Bs[brow * BK + swz(brow, ag) * 8]);
+            &Bs[brow * RP + ag * 8]);
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af, bf, acc[mi][ni], 0, 0, 0);
       }
@@ -141,9 +141,9 @@ void conv2d_fwd_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 __global__ __launch_bounds__(256)
 void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
                          const bf16* __restrict__ w, ConvParams p) {
-  constexpr int BM = 128, BN = 128, BK = 32;
-  __shared__ u16 As[BM * BK];
-  __shared__ u16 Bs[BN * BK];
+  constexpr int BM = 128, BN = 128, BK = 32, RP = 40;
+  __shared__ u16 As[BM * RP];
+  __shared__ u16 Bs[BN * RP];
 
   const int K = p.I * p.kh * p.kw;
   const int P = p.OH * p.OW;
@@ -180,7 +180,7 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
       int row = half * 64 + (t >> 2), g = t & 3;
       int o = m0 + row;
       int kk = k0 + g * 8;
-      u16* dst = &As[row * BK + (swz(row, g) << 3)];
+      u16* dst = &As[row * RP + (g << 3)];
       if (o < p.O && kk + 8 <= K && a_vec) {
         *reinterpret_cast<s16x8*>(dst) =
             *reinterpret_cast<const s16x8*>(wb + (long)o * K + kk);
@@ -213,8 +213,8 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
         if (vb && yb2 >= 0 && yb2 < p.H && xb2 >= 0 && xb2 < p.W)
           vb16 = __builtin_bit_cast(u16, xi[(long)yb2 * p.W + xb2]);
       }
-      Bs[lane * BK + (swz(lane, lk >> 3) << 3) + (lk & 7)] = va16;
-      Bs[(64 + lane) * BK + (swz(64 + lane, lk >> 3) << 3) + (lk & 7)] = vb16;
+      Bs[lane * RP + lk] = va16;
+      Bs[(64 + lane) * RP + lk] = vb16;
     }
     __syncthreads();
 #pragma unroll
@@ -222,12 +222,12 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
       int arow = wm + mi * 16 + (lane & 15);
       int ag = lane >> 4;
       s16x8 af = *reinterpret_cast<const s16x8*>(
-          &As[arow * BK + (swz(arow, ag) << 3)]);
+          &As[arow * RP + (ag << 3)]);
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         int brow = wn + ni * 16 + (lane & 15);
         s16x8 bfr = *reinterpret_cast<const s16x8*>(
-            &Bs[brow * BK + (swz(brow, ag) << 3)]);
+            &Bs[brow * RP + (ag << 3)]);
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af, bfr, acc[mi][ni], 0, 0, 0);
       }

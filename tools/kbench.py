@@ -1,0 +1,118 @@
+#!/usr/bin/env python3
+"""Microbenchmark for the hot HIP kernels (run on a GPU box).
+
+    python tools/kbench.py [conv|wgrad|attn|upfirdn|all]
+
+Prints per-shape wall time and achieved TFLOP/s (or GB/s for memory-bound
+ops) so kernel tuning can proceed without full-model noise.
+"""
+
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+
+import gansformer_amd._C as C  # noqa: E402
+
+
+def timeit(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.time() - t0) / iters
+
+
+def bench_conv(dtype=torch.bfloat16):
+    dev = "cuda:0"
+    shapes = [
+        # (B, I, O, H, k, stride, note) — flagship layer shapes
+        (8, 128, 128, 256, 3, 1, "res256 conv1"),
+        (8, 256, 128, 256, 3, 1, "res256 conv0(post-up)"),
+        (8, 256, 256, 128, 3, 1, "res128 conv1"),
+        (8, 512, 512, 64, 3, 1, "res64 conv"),
+        (8, 512, 512, 32, 3, 1, "res32 conv"),
+        (8, 512, 512, 16, 3, 1, "res16 conv"),
+        (8, 128, 3, 256, 1, 1, "tRGB"),
+        (8, 128, 256, 256, 3, 2, "D down 256->128"),
+    ]
+    for B, I, O, H, k, s, note in shapes:
+        x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
+        w = torch.randn(B, O, I, k, k, device=dev, dtype=dtype) * 0.05
+        OH = (H + 2 * (k // 2) - k) // s + 1
+        t = timeit(lambda: C.conv2d_fwd(x, w, s, k // 2))
+        fl = 2.0 * B * O * I * k * k * OH * OH
+        print(f"conv fwd {note:22s} B{B} {I:4d}->{O:4d} @{H:4d} k{k} s{s}: "
+              f"{t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
+
+
+def bench_wgrad(dtype=torch.bfloat16):
+    dev = "cuda:0"
+    shapes = [
+        (8, 128, 128, 256, 3, 1, "res256 conv1"),
+        (8, 512, 512, 64, 3, 1, "res64 conv"),
+        (8, 128, 3, 256, 1, 1, "tRGB"),
+        (8, 3, 128, 256, 1, 1, "fromRGB"),
+    ]
+    for B, I, O, H, k, s, note in shapes:
+        x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
+        OH = (H + 2 * (k // 2) - k) // s + 1
+        dy = torch.randn(B, O, OH, OH, device=dev, dtype=dtype)
+        t = timeit(lambda: C.conv2d_wgrad(x, dy, s, k // 2, k, k, True))
+        fl = 2.0 * B * O * I * k * k * OH * OH
+        print(f"wgrad    {note:22s} B{B} {I:4d}->{O:4d} @{H:4d}: "
+              f"{t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
+
+
+def bench_attn(dtype=torch.bfloat16):
+    dev = "cuda:0"
+    shapes = [
+        (8, 65536, 16, 128, 128, "simplex res256"),
+        (8, 16384, 16, 256, 256, "simplex res128"),
+        (8, 16, 16384, 256, 256, "duplex-rev res128"),
+        (8, 16, 65536, 128, 128, "duplex-rev res256"),
+    ]
+    for B, Nq, Nk, D, E, note in shapes:
+        q = torch.randn(B, Nq, D, device=dev, dtype=dtype)
+        k = torch.randn(B, Nk, D, device=dev, dtype=dtype)
+        v = torch.randn(B, Nk, E, device=dev, dtype=dtype)
+        t = timeit(lambda: C.bipartite_attn(q, k, v, D ** -0.5))
+        fl = 2.0 * B * Nq * Nk * (D + E)
+        print(f"attn     {note:22s}: {t * 1e3:8.3f} ms  "
+              f"{fl / t / 1e12:7.1f} TF")
+
+
+def bench_upfirdn(dtype=torch.bfloat16):
+    dev = "cuda:0"
+    from gansformer_amd.ops.upfirdn2d import setup_filter
+    f = setup_filter([1, 3, 3, 1], device=torch.device(dev))
+    shapes = [
+        (8, 128, 256, 1, 1, "blur res256"),
+        (8, 256, 128, 2, 1, "up2 res128->256"),
+        (8, 128, 256, 1, 2, "down2 res256->128"),
+    ]
+    for B, Cn, H, up, down, note in shapes:
+        x = torch.randn(B, Cn, H, H, device=dev, dtype=dtype)
+        t = timeit(lambda: C.upfirdn2d(x, f, up, up, down, down, 2, 1, 2, 1,
+                                       1.0))
+        nbytes = x.numel() * x.element_size() * (1 + up * up / (down * down))
+        print(f"upfirdn  {note:22s}: {t * 1e3:8.3f} ms  "
+              f"{nbytes / t / 1e9:7.1f} GB/s")
+
+
+if __name__ == "__main__":
+    which = sys.argv[1] if len(sys.argv) > 1 else "all"
+    torch.manual_seed(0)
+    if which in ("conv", "all"):
+        bench_conv()
+    if which in ("wgrad", "all"):
+        bench_wgrad()
+    if which in ("attn", "all"):
+        bench_attn()
+    if which in ("upfirdn", "all"):
+        bench_upfirdn()
