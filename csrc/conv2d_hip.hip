@@ -171,6 +171,14 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
   int ixb = (pix_b % p.OW) * p.stride - p.pad;
   const bool va = pix_a < P, vb = pix_b < P;
 
+  // incremental tap state (i, r, s) for this wave's first k of the step —
+  // advanced without division inside the K loop (divisions are VALU
+  // poison: kk/khw per lane per element made v2 VALU-bound).
+  int ti = (wave * 8) / khw;
+  int trs = (wave * 8) % khw;
+  int tr = trs / p.kw, ts = trs % p.kw;
+  const int q24 = 24 / khw, r24 = 24 % khw;  // advance by 24 = 32 - 8
+
   f32x4 acc[4][4] = {};
   const bool a_vec = (K % 8 == 0);
 
@@ -196,26 +204,40 @@ void conv2d_fwd_bf16_128(bf16* __restrict__ y, const bf16* __restrict__ x,
       }
     }
     // ---- stage B: 128 pixels x 32 k; wave w owns k range [w*8, w*8+8),
-    // lanes sweep pixels (coalesced 64-wide); tap decode is wave-uniform.
+    // lanes sweep pixels (coalesced 64-wide); incremental tap decode.
+    {
+      int i = ti, r = tr, s = ts;
 #pragma unroll
-    for (int kj = 0; kj < 8; ++kj) {
-      int lk = wave * 8 + kj;
-      int kk = k0 + lk;
-      u16 va16 = 0, vb16 = 0;
-      if (kk < K) {
-        int i = kk / khw;              // wave-uniform (SALU)
-        int rs = kk % khw;
-        int r = rs / p.kw, s = rs % p.kw;
-        const bf16* xi = xb + (long)i * p.H * p.W;
-        int ya = iya + r, xa = ixa + s;
-        if (va && ya >= 0 && ya < p.H && xa >= 0 && xa < p.W)
-          va16 = __builtin_bit_cast(u16, xi[(long)ya * p.W + xa]);
-        int yb2 = iyb + r, xb2 = ixb + s;
-        if (vb && yb2 >= 0 && yb2 < p.H && xb2 >= 0 && xb2 < p.W)
-          vb16 = __builtin_bit_cast(u16, xi[(long)yb2 * p.W + xb2]);
+      for (int kj = 0; kj < 8; ++kj) {
+        int lk = wave * 8 + kj;
+        int kk = k0 + lk;
+        u16 va16 = 0, vb16 = 0;
+        if (kk < K) {
+          const bf16* xi = xb + (long)i * p.H * p.W;
+          int ya = iya + r, xa = ixa + s;
+          if (va && ya >= 0 && ya < p.H && xa >= 0 && xa < p.W)
+            va16 = __builtin_bit_cast(u16, xi[(long)ya * p.W + xa]);
+          int yb2 = iyb + r, xb2 = ixb + s;
+          if (vb && yb2 >= 0 && yb2 < p.H && xb2 >= 0 && xb2 < p.W)
+            vb16 = __builtin_bit_cast(u16, xi[(long)yb2 * p.W + xb2]);
+        }
+        Bs[lane * RP + lk] = va16;
+        Bs[(64 + lane) * RP + lk] = vb16;
+        // advance (i, r, s) by one tap
+        if (++s == p.kw) {
+          s = 0;
+          if (++r == p.kh) { r = 0; ++i; }
+        }
       }
-      Bs[lane * RP + lk] = va16;
-      Bs[(64 + lane) * RP + lk] = vb16;
+      // advance the wave state by the remaining 24 to reach k0 + BK
+      ti = i + q24;
+      tr = r;
+      ts = s;
+      int rs = tr * p.kw + ts + r24;
+      if (rs >= khw) { rs -= khw; ++ti; }
+      tr = rs / 3;  // kw<=3 in this framework; exact for kw==3
+      ts = rs - tr * 3;
+      if (p.kw != 3) { tr = rs / p.kw; ts = rs % p.kw; }
     }
     __syncthreads();
 #pragma unroll
