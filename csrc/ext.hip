@@ -39,6 +39,11 @@ void launch_upfirdn2d(T*, const T*, const float*, const UfdParams&,
                       hipStream_t);
 void launch_conv2d_fwd_bf16(bf16*, const bf16*, const bf16*,
                             const ConvParams&, hipStream_t);
+bool conv2d_slab_eligible(int I, int O, int H, int W, int OH, int OW,
+                          int kh, int kw, int stride, int pad,
+                          int per_sample);
+void launch_conv2d_fwd_slab_bf16(bf16*, const bf16*, const bf16*, int B,
+                                 int I, int H, int W, int O, hipStream_t);
 void launch_conv2d_fwd_f32(float*, const float*, const float*,
                            const ConvParams&, hipStream_t);
 int conv2d_wgrad_nsplit(const ConvParams&);
@@ -175,6 +180,13 @@ Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
   if (x.scalar_type() == torch::kFloat32) {
     gfa::launch_conv2d_fwd_f32(ptr<float>(out), cptr<float>(x),
                                cptr<float>(w), p, cur_stream());
+  } else if (gfa::conv2d_slab_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
+                                       p.kw, p.stride, p.pad, p.per_sample)) {
+    // tap-major slab kernel wants weights as [O][kh*kw][I]
+    auto wr = w.permute({0, 2, 3, 1}).reshape({O, kh * kw, I}).contiguous();
+    gfa::launch_conv2d_fwd_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
+                                     cptr<gfa::bf16>(wr), p.B, p.I, p.H, p.W,
+                                     p.O, cur_stream());
   } else {
     gfa::launch_conv2d_fwd_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                 cptr<gfa::bf16>(w), p, cur_stream());

@@ -20,8 +20,8 @@ CSRC = os.path.dirname(os.path.abspath(__file__))
 REPO = os.path.dirname(CSRC)
 
 sources = [os.path.join(CSRC, f) for f in (
-    "ext.hip", "fba.hip", "upfirdn2d.hip", "conv2d.hip", "mbstd.hip",
-    "attn.hip")]
+    "ext.hip", "fba.hip", "upfirdn2d.hip", "conv2d.hip", "conv2d_slab.hip",
+    "mbstd.hip", "attn.hip")]
 
 setup(
     name="gansformer_amd_ext",

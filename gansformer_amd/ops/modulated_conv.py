@@ -2,10 +2,23 @@
 
 Capability parity with the reference's modulated conv (built in the TF
 graph around cuDNN grouped conv, ref src/training/networks.py [R],
-SURVEY.md K3). Here the per-sample modulated weight tensor [B,O,I,kh,kw]
-is fed directly to our per-sample implicit-GEMM conv kernel
-(conv2d_gradfix) — no grouped-conv reshaping trick.
+SURVEY.md K3), re-derived for MI355X memory behavior.
 
+Because convolution is linear in its input,
+
+    conv(x, w * s[b,i] * d[b,o])  ==  d[b,o] * conv(x * s[b,i], w)
+
+so instead of materializing a per-sample weight tensor [B,O,I,kh,kw]
+(151 MB/layer at 512ch/batch-32 — it made weights the dominant HBM
+traffic of the whole step), we scale the *input* by the styles, run a
+single SHARED-weight implicit-GEMM conv on MFMA (the base weight is
+4.7 MB and stays L2-resident across the whole grid), and apply the
+demodulation factor to the output. The demod factor itself is computed
+without materializing anything: sum_{i,kh,kw} (w*s)^2 over taps equals
+(s^2) @ W2^T with W2[o,i] = sum_{kh,kw} w^2 — one tiny [B,I]x[I,O] GEMM.
+
+All pieces are plain differentiable torch ops around conv2d_gradfix, so
+R1 / path-length double-backward is exact by construction.
 Modulation/demodulation math runs in fp32 (bf16-safe), the convolution
 runs in the activation dtype on the MFMA path.
 """
@@ -40,19 +53,26 @@ def modulated_conv2d(
         # pre-normalize for low-precision safety: bound |w|*|s| ~ 1
         w = w * w.square().mean([1, 2, 3], keepdim=True).rsqrt()
         s = s / s.abs().amax(dim=1, keepdim=True).clamp(min=1e-8)
-    w = w.unsqueeze(0) * s.reshape(B, 1, I, 1, 1)  # [B,O,I,kh,kw]
-    if demodulate:
-        d = (w.square().sum(dim=[2, 3, 4]) + 1e-8).rsqrt()  # [B,O]
-        w = w * d.reshape(B, O, 1, 1, 1)
+        # demod factor via the W2 trick: no [B,O,I,kh,kw] intermediate
+        w2 = w.square().sum(dim=[2, 3])            # [O, I]
+        d = (s.square() @ w2.t() + 1e-8).rsqrt()   # [B, O]
+
+    x = x * s.reshape(B, I, 1, 1).to(x.dtype)
     w = w.to(x.dtype)
 
     if up > 1:
         x = upsample2d(x, resample_filter, up=up)
-    if down > 1:
+        y = conv2d_gradfix(x, w, stride=1, padding=padding)
+    elif down > 1:
         # blur (same-size, with the downsample pad baked in), then strided conv
         fh = resample_filter.shape[0]
         p0 = (fh - down + 1) // 2 + padding
         p1 = (fh - down) // 2 + padding
         x = upfirdn2d(x, resample_filter, padding=(p0, p1, p0, p1))
-        return conv2d_gradfix(x, w, stride=down, padding=0)
-    return conv2d_gradfix(x, w, stride=1, padding=padding)
+        y = conv2d_gradfix(x, w, stride=down, padding=0)
+    else:
+        y = conv2d_gradfix(x, w, stride=1, padding=padding)
+
+    if demodulate:
+        y = y * d.reshape(B, O, 1, 1).to(y.dtype)
+    return y

@@ -96,6 +96,9 @@ def test_upfirdn2d_asym(dev):
     dict(I=3, O=32, H=16, k=1, stride=1, pad=0),     # fromRGB (K=3)
     dict(I=16, O=32, H=16, k=3, stride=2, pad=1),    # downsample conv
     dict(I=80, O=96, H=8, k=3, stride=1, pad=1),     # K not mult of 64
+    dict(I=32, O=48, H=16, k=3, stride=1, pad=1),    # slab path (bf16 shared)
+    dict(I=64, O=160, H=8, k=3, stride=1, pad=1),    # slab path, O tail > 128
+    dict(I=128, O=128, H=32, k=3, stride=1, pad=1),  # slab path, bigger
 ])
 def test_conv2d_fwd(dev, dtype, per_sample, cfg):
     torch.manual_seed(3)

@@ -40,10 +40,13 @@ def bench_conv(dtype=torch.bfloat16):
         (8, 512, 512, 16, 3, 1, "res16 conv"),
         (8, 128, 3, 256, 1, 1, "tRGB"),
         (8, 128, 256, 256, 3, 2, "D down 256->128"),
+        (32, 512, 512, 64, 3, 1, "res64 conv B32"),
+        (32, 256, 256, 128, 3, 1, "res128 conv1 B32"),
+        (32, 128, 128, 256, 3, 1, "res256 conv1 B32"),
     ]
     for B, I, O, H, k, s, note in shapes:
         x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
-        w = torch.randn(B, O, I, k, k, device=dev, dtype=dtype) * 0.05
+        w = torch.randn(O, I, k, k, device=dev, dtype=dtype) * 0.05
         OH = (H + 2 * (k // 2) - k) // s + 1
         t = timeit(lambda: C.conv2d_fwd(x, w, s, k // 2))
         fl = 2.0 * B * O * I * k * k * OH * OH
@@ -63,7 +66,7 @@ def bench_wgrad(dtype=torch.bfloat16):
         x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
         OH = (H + 2 * (k // 2) - k) // s + 1
         dy = torch.randn(B, O, OH, OH, device=dev, dtype=dtype)
-        t = timeit(lambda: C.conv2d_wgrad(x, dy, s, k // 2, k, k, True))
+        t = timeit(lambda: C.conv2d_wgrad(x, dy, s, k // 2, k, k, False))
         fl = 2.0 * B * O * I * k * k * OH * OH
         print(f"wgrad    {note:22s} B{B} {I:4d}->{O:4d} @{H:4d}: "
               f"{t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
