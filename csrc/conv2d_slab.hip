@@ -4,40 +4,42 @@
 // re-reads every input pixel up to 9 times from global memory; its
 // MFMA:staging ratio makes it staging-bound (~9% of bf16 MFMA peak,
 // profiles/r01_summary.md). This kernel reorders K as (tap, channel):
-// the 16x16-pixel output tile's input slab (18x18xBC, halo included) is
+// the 16x8-pixel output tile's input slab (18x10xBC, halo included) is
 // staged in LDS ONCE per 32-channel block, and the 9 taps' B-operand
 // fragments are read DIRECTLY from the slab at shifted offsets — no
-// per-tap global traffic, no per-tap LDS re-staging.
+// per-tap global traffic for B, no per-tap LDS re-staging.
 //
 //   C[o][p]  +=  sum_{t,c} W[o][t][c] * slab[(py+ty(t))(px+tx(t))][c]
 //
 // Weights are passed pre-permuted as [O][9][I] so A-fragments are
-// contiguous 16-B reads. A is double-buffered per tap; the slab for the
-// next channel block is prefetched during tap 0 (its latency hides
-// behind ~8 taps of MFMA).
+// contiguous 16-B reads. A-fragments are read straight from GLOBAL
+// memory (they are shared by every pixel tile and every sample, so they
+// live in L2: ~8 KB/tap/WG against a 34 TB/s aggregate L2). That leaves
+// the slab as the only LDS producer, so the tap loop runs with NO
+// barriers; one __syncthreads per 32-channel block. The next channel
+// block's slab is prefetched into the other buffer during the MFMA of
+// the current one.
 //
 // Tile: 128(O) x 128(pixels as 16 wide x 8 high), BC=32, 4 waves, each
 // wave a 64x64 sub-tile = 4x4 fragments of 16x16,
-// v_mfma_f32_16x16x32_bf16. Per wave per (channel-block, tap):
-// 8 ds_read_b128 + 16 MFMA, at 3 waves/SIMD occupancy.
+// v_mfma_f32_16x16x32_bf16, 3 waves/SIMD occupancy.
 //
 // Replaces the reference's cuDNN 3x3 convs (TF 1.14).
 #include "common.h"
+#include <type_traits>
 
 namespace gfa {
 
 namespace {
 constexpr int SLAB_PIX = 40;           // u16 per slab pixel (32 + 8 pad)
 constexpr int SLAB_N = 10 * 18 * SLAB_PIX;  // one slab buffer, u16
-constexpr int AROW = 40;               // u16 per A row (32 + 8 pad)
 }  // namespace
 
-__global__ __launch_bounds__(256, 3)
+__global__ __launch_bounds__(256, 2)
 void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
                           const bf16* __restrict__ wr,  // [O][9][I]
                           int B, int I, int H, int W, int O) {
   __shared__ u16 slab[2][SLAB_N];
-  __shared__ u16 As[2][128 * AROW];
 
   const int tilesX = W >> 4;
   const int b = blockIdx.z;
@@ -56,90 +58,95 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   const int px = lane & 15;               // B-fragment pixel x (fixed)
   const int ag = lane >> 4;               // K granule 0..3
 
-  // ---- staging roles ----
-  // slab: tasks (dy, c): idx -> dy = idx/32 in 0..17, c = idx%32.
-  // A: thread -> row t>>1 (0..127), half g2 = t&1 (16 u16 each).
-  const int a_row = t >> 1, a_g2 = t & 1;
-  const long wrow = (long)(m0 + a_row) * 9 * I + a_g2 * 16;
-  const bool a_ok = (m0 + a_row) < O;
+  // A-fragment global bases for this lane's 4 M rows (+tap*I+c0 later).
+  // Out-of-range O rows are CLAMPED, not masked: they compute garbage
+  // partial sums whose stores the epilogue skips, so no zero-fill or
+  // per-fragment predicate is needed (saves registers). Offsets fit
+  // int32 (O*9*I <= 2.4M elements).
+  int a_off[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    const int o = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
+    a_off[mi] = o * 9 * I + ag * 8;
+  }
 
   const int nCB = I >> 5;
 
-  // part < 0: whole slab; part 0..2: one 256-task slice (spread across
-  // taps so each tap segment's barrier only waits on 1/3 of the loads)
-  auto stage_slab = [&](int c0, int sb, int part) {
-    const int lo = part < 0 ? 0 : part * 160;
-    const int hi = part < 0 ? 10 * 32 : min(10 * 32, lo + 160);
-    for (int idx = lo + t; idx < hi; idx += 256) {
+  // Slab staging is split into a LOAD phase (global -> registers, issued
+  // at the top of a channel block) and a WRITE phase (registers -> LDS,
+  // after the tap loop): a combined load+write forces the s_waitcnt for
+  // HBM data before the block's first MFMA and parks the wave (~85%
+  // SQ_WAIT_ANY measured). Each thread owns <=2 (dy, c) row tasks of 24
+  // elements held as 6 s16x4.
+  const int n_task = (t < 320 - 256) ? 2 : 1;
+  s16x4 tk[2][6];
+
+  auto slab_load = [&](int c0) {
+#pragma unroll
+    for (int k = 0; k < 2; ++k) {
+      if (k >= n_task) break;
+      const int idx = t + k * 256;
       const int dy = idx >> 5, c = idx & 31;
       const int gy = y0 + dy - 1;
-      u16* dst = &slab[sb][(dy * 18) * SLAB_PIX + c];
       const bf16* src = xb + ((long)(c0 + c) * H + gy) * W;
       if (gy < 0 || gy >= H) {
 #pragma unroll
-        for (int dx = 0; dx < 18; ++dx) dst[dx * SLAB_PIX] = 0;
+        for (int j = 0; j < 6; ++j) tk[k][j] = s16x4{};
       } else if (x0 >= 4 && x0 + 20 <= W) {
         // interior: 6 aligned 8-B loads covering [x0-4, x0+20)
-        s16x4 r[6];
         const s16x4* sp = reinterpret_cast<const s16x4*>(src + x0 - 4);
 #pragma unroll
-        for (int j = 0; j < 6; ++j) r[j] = sp[j];
-        const u16* rp = reinterpret_cast<const u16*>(&r[0]);
-#pragma unroll
-        for (int dx = 0; dx < 18; ++dx) dst[dx * SLAB_PIX] = rp[3 + dx];
+        for (int j = 0; j < 6; ++j) tk[k][j] = sp[j];
       } else {
+        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
 #pragma unroll
-        for (int dx = 0; dx < 18; ++dx) {
-          const int gx = x0 - 1 + dx;
-          u16 v = 0;
-          if (gx >= 0 && gx < W)
-            v = __builtin_bit_cast(u16, src[gx]);
-          dst[dx * SLAB_PIX] = v;
+        for (int dx = 0; dx < 24; ++dx) {
+          const int gx = x0 - 4 + dx;
+          rp[dx] = (gx >= 0 && gx < W) ? __builtin_bit_cast(u16, src[gx])
+                                       : (u16)0;
         }
       }
     }
   };
 
-  auto stage_A = [&](int tap, int c0, int ab) {
-    u16* dst = &As[ab][a_row * AROW + a_g2 * 16];
-    if (a_ok) {
-      const s16x8* sp =
-          reinterpret_cast<const s16x8*>(wr + wrow + (long)tap * I + c0);
-      *reinterpret_cast<s16x8*>(dst) = sp[0];
-      *reinterpret_cast<s16x8*>(dst + 8) = sp[1];
-    } else {
-      *reinterpret_cast<s16x8*>(dst) = s16x8{};
-      *reinterpret_cast<s16x8*>(dst + 8) = s16x8{};
+  auto slab_write = [&](int sb) {
+#pragma unroll
+    for (int k = 0; k < 2; ++k) {
+      if (k >= n_task) break;
+      const int idx = t + k * 256;
+      const int dy = idx >> 5, c = idx & 31;
+      u16* dst = &slab[sb][(dy * 18) * SLAB_PIX + c];
+      const u16* rp = reinterpret_cast<const u16*>(&tk[k][0]);
+#pragma unroll
+      for (int dx = 0; dx < 18; ++dx) dst[dx * SLAB_PIX] = rp[3 + dx];
     }
   };
 
   f32x4 acc[4][4] = {};
 
-  stage_slab(0, 0, -1);
-  stage_A(0, 0, 0);
+  slab_load(0);
+  slab_write(0);
+
   __syncthreads();
 
+  // A-fragments load at use (L2-hot weights; 3 waves/SIMD hide each
+  // other's latency). Slab staging for the next channel block: loads
+  // issue at the top of the block, the LDS write lands after tap 1 so
+  // the register live-range of the staged rows stays short while the
+  // global latency still hides behind two taps of MFMA.
   for (int cb = 0; cb < nCB; ++cb) {
     const int c0 = cb << 5;
     const int sb = cb & 1;
+    const bool pre = cb + 1 < nCB;
+    if (pre) slab_load(c0 + 32);
 #pragma unroll
     for (int tap = 0; tap < 9; ++tap) {
-      // prefetch: next A buffer; during tap 0, also next channel slab
-      if (tap < 8) {
-        stage_A(tap + 1, c0, (tap + 1) & 1);
-      } else if (cb + 1 < nCB) {
-        stage_A(0, c0 + 32, 1);  // tap 9 ≡ buffer (9)&1 = 1
-      }
-      if (tap < 2 && cb + 1 < nCB) stage_slab(c0 + 32, sb ^ 1, tap);
-
       const int r = tap / 3, s = tap - r * 3;
-      const u16* ab = &As[tap & 1][0];
-      // A fragments for this wave (4), B fragments (8), 32 MFMA
-      s16x8 afr[4];
+      const int tc = tap * I + c0;
+      s16x8 af[4];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
-        afr[mi] = *reinterpret_cast<const s16x8*>(
-            &ab[(wm + mi * 16 + (lane & 15)) * AROW + (ag << 3)]);
+        af[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + tc);
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int py = wpy + ni;
@@ -148,10 +155,11 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[mi], bfr, acc[mi][ni], 0, 0, 0);
+              af[mi], bfr, acc[mi][ni], 0, 0, 0);
       }
-      __syncthreads();
+      if (tap == 1 && pre) slab_write(sb ^ 1);
     }
+    __syncthreads();
   }
 
   // ---- epilogue: scalar bf16 stores (stride H*W between o rows) ----

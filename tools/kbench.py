@@ -108,9 +108,22 @@ def bench_upfirdn(dtype=torch.bfloat16):
               f"{nbytes / t / 1e9:7.1f} GB/s")
 
 
+def bench_conv_one(dtype=torch.bfloat16):
+    # single hot shape, many iters — for rocprofv3 --pmc runs
+    dev = "cuda:0"
+    B, I, O, H, k, s = 32, 512, 512, 64, 3, 1
+    x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
+    w = torch.randn(O, I, k, k, device=dev, dtype=dtype) * 0.05
+    t = timeit(lambda: C.conv2d_fwd(x, w, s, k // 2), iters=30, warmup=5)
+    fl = 2.0 * B * O * I * k * k * H * H
+    print(f"conv fwd res64 B32: {t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     torch.manual_seed(0)
+    if which == "conv1":
+        bench_conv_one()
     if which in ("conv", "all"):
         bench_conv()
     if which in ("wgrad", "all"):
