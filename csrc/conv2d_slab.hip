@@ -97,6 +97,29 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
         const s16x4* sp = reinterpret_cast<const s16x4*>(src + x0 - 4);
 #pragma unroll
         for (int j = 0; j < 6; ++j) tk[k][j] = sp[j];
+      } else if (x0 == 0 && W >= 24) {
+        // left edge: aligned loads from 0; slot 3 (x = -1) zeroed in
+        // the write phase never moves — shift source window one slot
+        const s16x4* sp = reinterpret_cast<const s16x4*>(src);
+        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
+        s16x4 tmp[5];
+#pragma unroll
+        for (int j = 0; j < 5; ++j) tmp[j] = sp[j];
+        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
+        rp[3] = 0;  // x = -1
+#pragma unroll
+        for (int dx = 1; dx < 18; ++dx) rp[3 + dx] = tp[dx - 1];
+      } else if (x0 + 16 == W && W >= 24) {
+        // right edge: aligned loads from W-24; x = x0-1+dx = (W-24)+7+dx
+        const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 24);
+        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
+        s16x4 tmp[6];
+#pragma unroll
+        for (int j = 0; j < 6; ++j) tmp[j] = sp[j];
+        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
+#pragma unroll
+        for (int dx = 0; dx < 17; ++dx) rp[3 + dx] = tp[7 + dx];
+        rp[3 + 17] = 0;  // x = W
       } else {
         u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
 #pragma unroll
