@@ -1,0 +1,229 @@
+// Tap-major weight-gradient kernel for 3x3 stride-1 convs on gfx950.
+//
+// dw[o][r][s][c] = sum_{b,p} dy[b][o][p] * x[b][c][p + (r-1, s-1)]
+//
+// The generic wgrad in conv2d.hip re-stages dY and x patches per 64x64
+// output tile with 2x2 fragments/wave — staging-bound. Here:
+//
+//  * K order is pixels, in 32-wide row chunks; each k-step's x window
+//    (32 ch x 3 rows x 34 cols) is loaded from global ONCE and expanded
+//    in LDS into its 9 tap-shifted copies (an im2col image in LDS: the
+//    9x read amplification happens on LDS bandwidth, not HBM), so every
+//    B fragment is an aligned 16-B ds_read.
+//  * dY (the A operand) is read directly from global. Blocks are
+//    swizzled so all I/32 channel-tiles of one (O-tile, K-split) land on
+//    the SAME XCD (blockIdx%8 picks the XCD): they stream the same dY
+//    slice through that XCD's L2, so dY's HBM traffic stays ~1x instead
+//    of x(I/32).
+//  * One block computes dw for 128 O x 32 C x all 9 taps; per wave
+//    4(M) x 1(C) x 9(tap) accumulators, 36 MFMA per 32-pixel k-step
+//    against 4 A-fragment loads. f32 atomics accumulate across K-splits.
+//
+// Replaces the cuDNN/TF wgrad of the reference.
+#include "common.h"
+
+namespace gfa {
+
+namespace {
+constexpr int SLABX = 40;               // u16 per (c, r, s) pixel row
+constexpr int SLAB_N = 32 * 3 * 3 * SLABX;  // one buffer: c, r, s, px
+}  // namespace
+
+__global__ __launch_bounds__(256, 2)
+void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
+                            const bf16* __restrict__ x,
+                            const bf16* __restrict__ dy,
+                            int B, int I, int H, int W, int O, int nsplit) {
+  __shared__ u16 slab[2][SLAB_N];
+
+  // block swizzle: bx = xcd + 8*(ct + nCt*pairHi); pair = xcd + 8*pairHi
+  const int nCt = I >> 5;
+  const int bx = blockIdx.x;
+  const int xcd = bx & 7;
+  const int rest = bx >> 3;
+  const int ct = rest % nCt;
+  const int pair = xcd + 8 * (rest / nCt);
+  const int nMt = (O + 127) >> 7;
+  const int m0 = (pair % nMt) << 7;
+  const int split = pair / nMt;
+  const int c0 = ct << 5;
+  if (pair >= nMt * nsplit) return;  // swizzle padding
+
+  const int OWc = W >> 5;                     // 32-px chunks per row
+  const long iters = (long)B * H * OWc;
+  const long span = (iters + nsplit - 1) / nsplit;
+  const long it0 = (long)split * span;
+  const long it1 = min(iters, it0 + span);
+  if (it0 >= it1) return;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 64;            // O offset of this wave
+  const int wc = (wave & 1) * 16;             // C offset of this wave
+  const int ag = lane >> 4;                   // pixel granule 0..3
+
+  // A (dY) fragment bases: 4 M rows per wave, this lane's row+granule
+  int a_row[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    a_row[mi] = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
+
+  // staging: 96 (c, dr) row tasks; threads 0..95 own one each,
+  // holding the 40-elem x window [col0-4, col0+36) as 10 s16x4.
+  const int s_c = t & 31, s_dr = t >> 5;      // dr 0..2 for t < 96
+  const bool stager = t < 96;
+  s16x4 tk[10];
+
+  auto win_load = [&](long it) {
+    const int b = (int)(it / ((long)H * OWc));
+    const int rem = (int)(it - (long)b * H * OWc);
+    const int row = rem / OWc;
+    const int col0 = (rem - row * OWc) << 5;
+    const int gy = row + s_dr - 1;
+    const bf16* src = x + (((long)b * I + c0 + s_c) * H + gy) * W;
+    if (!stager) return;
+    if (gy < 0 || gy >= H) {
+#pragma unroll
+      for (int j = 0; j < 10; ++j) tk[j] = s16x4{};
+    } else if (col0 >= 4 && col0 + 36 <= W) {
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src + col0 - 4);
+#pragma unroll
+      for (int j = 0; j < 10; ++j) tk[j] = sp[j];
+    } else if (W == 32) {
+      // single-chunk rows: load [0, 32), zeros off both ends
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src);
+      u16* rp = reinterpret_cast<u16*>(&tk[0]);
+      s16x4 tmp[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) tmp[j] = sp[j];
+      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
+      rp[3] = 0;
+#pragma unroll
+      for (int i = 0; i < 32; ++i) rp[4 + i] = tp[i];
+      rp[36] = 0;
+    } else if (col0 == 0) {
+      // left edge: loads [0, 36), x=-1 is zero
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src);
+      u16* rp = reinterpret_cast<u16*>(&tk[0]);
+      s16x4 tmp[9];
+#pragma unroll
+      for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
+      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
+      rp[3] = 0;
+#pragma unroll
+      for (int i = 0; i < 33; ++i) rp[4 + i] = tp[i];
+    } else {
+      // right edge: col0 + 32 == W, loads [W-36, W), x=W is zero
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 36);
+      u16* rp = reinterpret_cast<u16*>(&tk[0]);
+      s16x4 tmp[9];
+#pragma unroll
+      for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
+      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
+#pragma unroll
+      for (int i = 0; i < 33; ++i) rp[3 + i] = tp[3 + i];
+      rp[36] = 0;
+    }
+  };
+
+  // expand the register window into the 3 tap-shifted LDS copies:
+  // copy s holds pixels [col0 + s - 1, col0 + s + 31) at 16-B alignment
+  auto win_write = [&](int sb) {
+    if (!stager) return;
+    const u16* rp = reinterpret_cast<const u16*>(&tk[0]);
+    u16* base = &slab[sb][(s_c * 3 + s_dr) * 3 * SLABX];
+#pragma unroll
+    for (int s = 0; s < 3; ++s) {
+      u16* dst = base + s * SLABX;
+#pragma unroll
+      for (int i = 0; i < 32; ++i) dst[i] = rp[3 + s + i];
+    }
+  };
+
+  f32x4 acc[9][4] = {};
+
+  win_load(it0);
+  win_write(0);
+  __syncthreads();
+
+  for (long it = it0; it < it1; ++it) {
+    const int sb = (int)((it - it0) & 1);
+    if (it + 1 < it1) win_load(it + 1);
+
+    // A fragments for this k-step (32 dY pixels starting at this chunk)
+    const int b = (int)(it / ((long)H * OWc));
+    const int rem = (int)(it - (long)b * H * OWc);
+    const long dybase = ((long)b * O) * H * W + rem * 32 + ag * 8;
+    s16x8 af[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      af[mi] = *reinterpret_cast<const s16x8*>(
+          dy + dybase + (long)a_row[mi] * H * W);
+
+    const u16* sl = &slab[sb][0];
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      const int r = tap / 3, s = tap - r * 3;
+      // B fragment: 16 channels (lane&15) x 8 pixels (granule)
+      const s16x8 bfr = *reinterpret_cast<const s16x8*>(
+          &sl[(((wc + (lane & 15)) * 3 + r) * 3 + s) * SLABX + (ag << 3)]);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        acc[tap][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bfr, acc[tap][mi], 0, 0, 0);
+    }
+    if (it + 1 < it1) win_write(sb ^ 1);
+    __syncthreads();
+  }
+
+  // epilogue: C rows = O, cols = channels; atomic f32 accumulate
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+    const int c = c0 + wc + (lane & 15);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int o0 = m0 + wm + mi * 16 + ag * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int o = o0 + reg;
+        if (o < O)
+          atomicAdd(&ws[((long)o * 9 + tap) * I + c], acc[tap][mi][reg]);
+      }
+    }
+  }
+}
+
+bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
+                                int kh, int kw, int stride, int pad,
+                                int per_sample) {
+  return !per_sample && kh == 3 && kw == 3 && stride == 1 && pad == 1 &&
+         (I & 31) == 0 && I >= 32 && (W & 31) == 0 && W >= 32 &&
+         OH == H && OW == W;
+}
+
+int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O) {
+  const int nCt = I >> 5;
+  const int nMt = (O + 127) >> 7;
+  const long iters = (long)B * H * (W >> 5);
+  int nsplit = (int)((1024 + (long)nCt * nMt - 1) / ((long)nCt * nMt));
+  if (nsplit > iters) nsplit = (int)iters;
+  if (nsplit > 64) nsplit = 64;
+  if (nsplit < 1) nsplit = 1;
+  return nsplit;
+}
+
+void launch_conv2d_wgrad_slab_bf16(float* ws, const bf16* x, const bf16* dy,
+                                   int B, int I, int H, int W, int O,
+                                   int nsplit, hipStream_t s) {
+  const int nCt = I >> 5;
+  const int nMt = (O + 127) >> 7;
+  // pair ids run 0..nMt*nsplit-1; grid.x covers xcd-slot * ct * pairHi
+  const int npair = nMt * nsplit;
+  const int pair_hi = (npair + 7) >> 3;
+  dim3 grid((unsigned)(8 * nCt * pair_hi));
+  hipLaunchKernelGGL(conv2d_wgrad_slab_bf16, grid, dim3(256), 0, s, ws, x,
+                     dy, B, I, H, W, O, nsplit);
+}
+
+}  // namespace gfa

@@ -44,6 +44,13 @@ bool conv2d_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                           int per_sample);
 void launch_conv2d_fwd_slab_bf16(bf16*, const bf16*, const bf16*, int B,
                                  int I, int H, int W, int O, hipStream_t);
+bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
+                                int kh, int kw, int stride, int pad,
+                                int per_sample);
+int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O);
+void launch_conv2d_wgrad_slab_bf16(float*, const bf16*, const bf16*, int B,
+                                   int I, int H, int W, int O, int nsplit,
+                                   hipStream_t);
 void launch_conv2d_fwd_f32(float*, const float*, const float*,
                            const ConvParams&, hipStream_t);
 int conv2d_wgrad_nsplit(const ConvParams&);
@@ -205,6 +212,22 @@ Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
                        per_sample);
   TORCH_CHECK(p.OH == dy.size(2) && p.OW == dy.size(3),
               "conv2d_wgrad: dy shape mismatch");
+  if (x.scalar_type() == torch::kBFloat16 &&
+      gfa::conv2d_wgrad_slab_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
+                                      p.kw, p.stride, p.pad, p.per_sample)) {
+    // tap-major kernel accumulates f32 into [O][9][I]; rearrange after
+    auto ws = torch::zeros({O, 9, x.size(1)},
+                           x.options().dtype(torch::kFloat32));
+    int ns = gfa::conv2d_wgrad_slab_nsplit(p.B, p.I, p.H, p.W, p.O);
+    gfa::launch_conv2d_wgrad_slab_bf16(ws.data_ptr<float>(),
+                                       cptr<gfa::bf16>(x),
+                                       cptr<gfa::bf16>(dy), p.B, p.I, p.H,
+                                       p.W, p.O, ns, cur_stream());
+    return ws.reshape({O, 3, 3, x.size(1)})
+        .permute({0, 3, 1, 2})
+        .contiguous()
+        .to(x.scalar_type());
+  }
   auto shape = per_sample
                    ? std::vector<int64_t>{x.size(0), O, x.size(1), kh, kw}
                    : std::vector<int64_t>{O, x.size(1), kh, kw};

@@ -61,6 +61,9 @@ def bench_wgrad(dtype=torch.bfloat16):
         (8, 512, 512, 64, 3, 1, "res64 conv"),
         (8, 128, 3, 256, 1, 1, "tRGB"),
         (8, 3, 128, 256, 1, 1, "fromRGB"),
+        (32, 512, 512, 64, 3, 1, "res64 conv B32"),
+        (32, 256, 256, 128, 3, 1, "res128 conv1 B32"),
+        (32, 128, 128, 256, 3, 1, "res256 conv1 B32"),
     ]
     for B, I, O, H, k, s, note in shapes:
         x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
