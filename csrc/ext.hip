@@ -44,6 +44,10 @@ bool conv2d_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                           int per_sample);
 void launch_conv2d_fwd_slab_bf16(bf16*, const bf16*, const bf16*, int B,
                                  int I, int H, int W, int O, hipStream_t);
+bool conv2d_up2_eligible(int I, int O, int H, int W, int kh, int kw,
+                         int per_sample);
+void launch_conv2d_up2_slab_bf16(bf16*, const bf16*, const bf16*, int B,
+                                 int I, int H, int W, int O, hipStream_t);
 bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                                 int kh, int kw, int stride, int pad,
                                 int per_sample);
@@ -201,6 +205,28 @@ Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
   return out;
 }
 
+Tensor conv2d_up2(Tensor x, Tensor w) {
+  // y = conv2d(zero_stuff2(x), w, pad=1): parity-decomposed transposed
+  // conv at input resolution (1/4 the MACs of conv at output res)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() && w.dim() == 4);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16,
+              "conv2d_up2: bf16 only");
+  const int B = (int)x.size(0), I = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int O = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == I && w.size(2) == 3 && w.size(3) == 3);
+  TORCH_CHECK(gfa::conv2d_up2_eligible(I, O, H, W, 3, 3, 0),
+              "conv2d_up2: shape not eligible (I%32, H%8, W%16)");
+  auto wr = w.permute({0, 2, 3, 1}).reshape({O, 9, I}).contiguous();
+  auto out = torch::empty({B, O, 2 * H, 2 * W}, x.options());
+  gfa::launch_conv2d_up2_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
+                                   cptr<gfa::bf16>(wr), B, I, H, W, O,
+                                   cur_stream());
+  return out;
+}
+
 Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
                     int64_t kh, int64_t kw, bool per_sample) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
@@ -316,6 +342,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upfirdn2d", &upfirdn2d, "pad-upsample-FIR-downsample");
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv2d (per-sample ok)");
   m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight gradient");
+  m.def("conv2d_up2", &conv2d_up2,
+        "2x-upsampling 3x3 conv (= conv of zero-stuffed input, pad 1)");
   m.def("mbstd", &mbstd, "minibatch stddev stats [B/G, F]");
   m.def("bipartite_attn", &bipartite_attn, "softmax(QK^T)V");
 }

@@ -27,7 +27,7 @@ from __future__ import annotations
 
 import torch
 
-from .conv2d_grad import conv2d_gradfix
+from .conv2d_grad import conv2d_gradfix, conv2d_up2
 from .upfirdn2d import upsample2d, upfirdn2d
 
 
@@ -61,8 +61,15 @@ def modulated_conv2d(
     w = w.to(x.dtype)
 
     if up > 1:
-        x = upsample2d(x, resample_filter, up=up)
-        y = conv2d_gradfix(x, w, stride=1, padding=padding)
+        # transposed conv at input resolution (1/4 the MACs of
+        # upsample-then-conv), then the FIR blur — convs commute, so in
+        # the interior this equals blurring first like the reference did
+        assert up == 2 and kh == 3 and padding == 1
+        t = conv2d_up2(x, w)
+        fh = resample_filter.shape[0]
+        y = upfirdn2d(t, resample_filter,
+                      padding=(fh // 2, fh // 2 - 1, fh // 2, fh // 2 - 1),
+                      gain=up * up)
     elif down > 1:
         # blur (same-size, with the downsample pad baked in), then strided conv
         fh = resample_filter.shape[0]

@@ -253,3 +253,22 @@ def test_r1_double_backward_gpu(dev):
     realc = real.detach().cpu().requires_grad_(True)
     r1c = r1_penalty(Dc(realc), realc)
     assert abs(r1.item() - r1c.item()) / (abs(r1c.item()) + 1e-6) < 1e-3
+
+
+@pytest.mark.parametrize("cfg", [
+    dict(B=2, I=32, O=48, H=16),
+    dict(B=2, I=64, O=96, H=16),
+    dict(B=1, I=128, O=128, H=32),
+])
+def test_conv2d_up2(dev, cfg):
+    """Parity-decomposed up2 conv kernel vs eager zero-stuff reference."""
+    from gansformer_amd.ops.conv2d_grad import _zero_stuff2
+    import torch.nn.functional as F
+    torch.manual_seed(6)
+    B, I, O, H = cfg["B"], cfg["I"], cfg["O"], cfg["H"]
+    x = torch.randn(B, I, H, H, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(O, I, 3, 3, device=dev, dtype=torch.bfloat16) * 0.1
+    y = _C().conv2d_up2(x, w)
+    ref = F.conv2d(_zero_stuff2(x.float().cpu()), w.float().cpu(), padding=1)
+    assert y.shape == ref.shape
+    close(y, ref, torch.bfloat16)

@@ -156,3 +156,24 @@ def test_bipartite_attention_gradcheck():
                      eps=1e-6, atol=1e-4)
     assert gradgradcheck(lambda *t: bipartite_attention(*t), (q, k, v),
                          eps=1e-6, atol=1e-4)
+
+
+def test_conv2d_up2_matches_zero_stuff():
+    from gansformer_amd.ops.conv2d_grad import conv2d_up2, _zero_stuff2
+    import torch.nn.functional as F
+    torch.manual_seed(11)
+    x = torch.randn(2, 4, 6, 6, dtype=torch.float64)
+    w = torch.randn(5, 4, 3, 3, dtype=torch.float64) * 0.2
+    y = conv2d_up2(x, w)
+    ref = F.conv2d(_zero_stuff2(x), w, padding=1)
+    assert y.shape == (2, 5, 12, 12)
+    assert torch.allclose(y, ref, atol=1e-12)
+
+
+def test_conv2d_up2_gradcheck():
+    from gansformer_amd.ops.conv2d_grad import conv2d_up2
+    torch.manual_seed(12)
+    x = torch.randn(1, 3, 4, 4, dtype=torch.float64, requires_grad=True)
+    w = torch.randn(2, 3, 3, 3, dtype=torch.float64, requires_grad=True) * 0.3
+    assert gradcheck(conv2d_up2, (x, w), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(conv2d_up2, (x, w), eps=1e-6, atol=1e-4)
