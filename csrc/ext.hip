@@ -44,6 +44,9 @@ bool conv2d_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                           int per_sample);
 void launch_conv2d_fwd_slab_bf16(bf16*, const bf16*, const bf16*, int B,
                                  int I, int H, int W, int O, hipStream_t);
+void launch_upfirdn2d_sep4_bf16(bf16*, const bf16*, const float*, int, int,
+                                int, int, int, int, int, int, int, int,
+                                float, hipStream_t);
 bool conv2d_up2_eligible(int I, int O, int H, int W, int kh, int kw,
                          int per_sample);
 void launch_conv2d_up2_slab_bf16(bf16*, const bf16*, const bf16*, int B,
@@ -157,6 +160,29 @@ Tensor upfirdn2d(Tensor x, Tensor f, int64_t upx, int64_t upy, int64_t downx,
     gfa::launch_upfirdn2d<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(x),
                                     f.data_ptr<float>(), p, cur_stream());
   });
+  return out;
+}
+
+Tensor upfirdn2d_sep(Tensor x, Tensor f8, int64_t u, int64_t d, int64_t px0,
+                     int64_t px1, int64_t py0, int64_t py1, double gain) {
+  // separable 4-tap path: f8 = concat(fy[4], fx[4]); u/d symmetric,
+  // (u,d) in {(1,1),(2,1),(1,2)}; bf16 only
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "upfirdn2d_sep: bf16");
+  TORCH_CHECK(f8.is_cuda() && f8.is_contiguous() && f8.numel() == 8 &&
+              f8.scalar_type() == torch::kFloat32);
+  TORCH_CHECK((u == 1 && d == 1) || (u == 2 && d == 1) || (u == 1 && d == 2),
+              "upfirdn2d_sep: unsupported up/down");
+  const int B = (int)x.size(0), C = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int OH = (int)((H * u + py0 + py1 - 4) / d + 1);
+  const int OW = (int)((W * u + px0 + px1 - 4) / d + 1);
+  TORCH_CHECK(OH > 0 && OW > 0, "upfirdn2d_sep: empty output");
+  auto out = torch::empty({B, C, OH, OW}, x.options());
+  gfa::launch_upfirdn2d_sep4_bf16(
+      ptr<gfa::bf16>(out), cptr<gfa::bf16>(x), f8.data_ptr<float>(), B, C, H,
+      W, OH, OW, (int)u, (int)d, (int)px0, (int)py0, (float)gain,
+      cur_stream());
   return out;
 }
 
@@ -340,6 +366,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "gansformer_amd gfx950 HIP kernels";
   m.def("fba", &fba, "fused bias+act (grad=0) / grad variant (grad=1)");
   m.def("upfirdn2d", &upfirdn2d, "pad-upsample-FIR-downsample");
+  m.def("upfirdn2d_sep", &upfirdn2d_sep,
+        "separable 4-tap upfirdn (bf16, two-pass LDS)");
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv2d (per-sample ok)");
   m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight gradient");
   m.def("conv2d_up2", &conv2d_up2,

@@ -1,0 +1,159 @@
+// Separable 4-tap upfirdn2d for the three hot resample cases:
+//   (up=1,down=1)  same-size FIR blur
+//   (up=2,down=1)  zero-stuff 2x upsample + blur
+//   (up=1,down=2)  blur + 2x downsample
+//
+// The flagship filter [1,3,3,1] (and every setup_filter() of a 1-D spec)
+// is rank-1, so the 2D FIR factors into a horizontal then a vertical
+// 4-tap pass. Per output that is 8 MACs instead of 16, and both passes
+// run out of LDS staged once per tile; stores are 16-B vectorized.
+// Replaces the generic/tiled path of upfirdn2d.hip for these cases
+// (ref upfirdn_2d.cu semantics).
+#include "common.h"
+
+namespace gfa {
+
+struct UfdParams;  // defined in upfirdn2d.hip; sep kernels take scalars
+
+// Tile geometry per variant: OUT tile TOWxTOH, input patch IPWxIPH.
+//   U=1,D=1: out 64x32, in 67x35
+//   U=2,D=1: out 64x32, in (66/2+2)=35 x (34/2+2)=19
+//   U=1,D=2: out 32x16, in 67x35
+template <typename T, int U, int D>
+__global__ __launch_bounds__(256)
+void upfirdn2d_sep4(T* __restrict__ out, const T* __restrict__ x,
+                    const float* __restrict__ f4,  // fy[4] then fx[4]
+                    int B, int C, int H, int W, int OH, int OW,
+                    int px0, int py0, float gain) {
+  constexpr int TOW = (D == 2) ? 32 : 64;
+  constexpr int TOH = (D == 2) ? 16 : 32;
+  constexpr int IPW = ((TOW - 1) * D + 3) / U + 2;
+  constexpr int IPH = ((TOH - 1) * D + 3) / U + 2;
+  constexpr int IPWP = IPW + (IPW % 2 ? 1 : 2);   // pad rows (bank spread)
+  __shared__ float sIn[IPH * IPWP];
+  __shared__ float sHb[IPH * (TOW + 4)];          // h-filtered, raw rows
+  constexpr int HBW = TOW + 4;
+
+  const float fy0 = f4[0], fy1 = f4[1], fy2 = f4[2], fy3 = f4[3];
+  const float fx0 = f4[4], fx1 = f4[5], fx2 = f4[6], fx3 = f4[7];
+
+  const int t = threadIdx.x;
+  const int tiles_x = (OW + TOW - 1) / TOW;
+  const int tiles_y = (OH + TOH - 1) / TOH;
+  const long ntiles = (long)tiles_x * tiles_y * B * C;
+
+  for (long tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int tx = (int)(tile % tiles_x);
+    const int ty = (int)((tile / tiles_x) % tiles_y);
+    const long bc = tile / ((long)tiles_x * tiles_y);
+    const T* xp = x + bc * H * W;
+    const int ox0 = tx * TOW, oy0 = ty * TOH;
+    // first input row/col the tile touches (z = o*D - p, i = floor(z/U))
+    const int zx0 = ox0 * D - px0, zy0 = oy0 * D - py0;
+    const int ix0 = (zx0 >= 0) ? zx0 / U : -((-zx0 + U - 1) / U);
+    const int iy0 = (zy0 >= 0) ? zy0 / U : -((-zy0 + U - 1) / U);
+
+    // ---- stage raw input patch (elementwise: coalesced across lanes)
+    for (int i = t; i < IPH * IPW; i += 256) {
+      const int r = i / IPW, c = i - r * IPW;
+      const int iy = iy0 + r, ix = ix0 + c;
+      float v = 0.f;
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+        v = to_f32(xp[(long)iy * W + ix]);
+      sIn[r * IPWP + c] = v;
+    }
+    __syncthreads();
+
+    // ---- horizontal pass: raw rows x TOW output columns
+    for (int i = t; i < IPH * TOW; i += 256) {
+      const int r = i / TOW, cx = i - r * TOW;
+      const int ox = ox0 + cx;
+      float acc = 0.f;
+      if (U == 1) {
+        const int c = ox * D - px0 - ix0;  // tap tx=0 position in patch
+        const float* row = &sIn[r * IPWP + c];
+        acc = row[0] * fx3 + row[1] * fx2 + row[2] * fx1 + row[3] * fx0;
+      } else {
+        // U==2: taps hit real samples only when (zx & 1) == 0
+        const int zxb = ox - px0;           // D==1
+        const int e = zxb & 1;
+        const int c = ((zxb + e) >> 1) - ix0;  // first valid sample >= zxb
+        const float* row = &sIn[r * IPWP + c];
+        if (e) {  // odd phase: taps 1,3 -> f[2], f[0]
+          acc = row[0] * fx2 + row[1] * fx0;
+        } else {  // even phase: taps 0,2 -> f[3], f[1]
+          acc = row[0] * fx3 + row[1] * fx1;
+        }
+      }
+      sHb[r * HBW + cx] = acc;
+    }
+    __syncthreads();
+
+    // ---- vertical pass + vectorized store: 8 outputs per thread
+    for (int i = t; i < TOH * TOW / 8; i += 256) {
+      const int row8 = TOW / 8;
+      const int cy = i / row8, cx8 = (i - cy * row8) * 8;
+      const int oy = oy0 + cy;
+      if (oy >= OH) continue;
+      typename Raw<T>::type vals[8];
+      const int zyb = oy * D - py0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int cx = cx8 + j;
+        float acc;
+        if (U == 1) {
+          const int r = zyb - iy0;
+          const float* col = &sHb[r * HBW + cx];
+          acc = col[0] * fy3 + col[HBW] * fy2 + col[2 * HBW] * fy1 +
+                col[3 * HBW] * fy0;
+        } else {
+          const int e = zyb & 1;
+          const int r = ((zyb + e) >> 1) - iy0;
+          const float* col = &sHb[r * HBW + cx];
+          acc = e ? (col[0] * fy2 + col[HBW] * fy0)
+                  : (col[0] * fy3 + col[HBW] * fy1);
+        }
+        vals[j] = f32_to_raw<typename Raw<T>::type>(acc * gain);
+      }
+      const int ox = ox0 + cx8;
+      // tiles are full except at the right/bottom edge of odd sizes
+      if (ox + 8 <= OW) {
+        *reinterpret_cast<s16x8*>(
+            reinterpret_cast<typename Raw<T>::type*>(out) +
+            bc * (long)OH * OW + (long)oy * OW + ox) =
+            *reinterpret_cast<const s16x8*>(&vals[0]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (ox + j < OW)
+            reinterpret_cast<typename Raw<T>::type*>(out)
+                [bc * (long)OH * OW + (long)oy * OW + ox + j] = vals[j];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// bf16-only: the s16x8 vector store assumes a 2-byte element; fp32
+// traffic goes through the generic/tiled kernels in upfirdn2d.hip.
+void launch_upfirdn2d_sep4_bf16(bf16* out, const bf16* x, const float* f4,
+                                int B, int C, int H, int W, int OH, int OW,
+                                int U, int D, int px0, int py0, float gain,
+                                hipStream_t s) {
+  const int TOW = (D == 2) ? 32 : 64;
+  const int TOH = (D == 2) ? 16 : 32;
+  long ntiles = (long)((OW + TOW - 1) / TOW) * ((OH + TOH - 1) / TOH) * B * C;
+  if (ntiles > 8192) ntiles = 8192;
+  dim3 grid((unsigned)ntiles);
+  if (U == 1 && D == 1)
+    hipLaunchKernelGGL((upfirdn2d_sep4<bf16, 1, 1>), grid, dim3(256), 0, s,
+                       out, x, f4, B, C, H, W, OH, OW, px0, py0, gain);
+  else if (U == 2 && D == 1)
+    hipLaunchKernelGGL((upfirdn2d_sep4<bf16, 2, 1>), grid, dim3(256), 0, s,
+                       out, x, f4, B, C, H, W, OH, OW, px0, py0, gain);
+  else
+    hipLaunchKernelGGL((upfirdn2d_sep4<bf16, 1, 2>), grid, dim3(256), 0, s,
+                       out, x, f4, B, C, H, W, OH, OW, px0, py0, gain);
+}
+
+}  // namespace gfa

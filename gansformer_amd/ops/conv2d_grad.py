@@ -152,13 +152,17 @@ class _Conv2dUp2(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w):
         ctx.save_for_backward(x, w)
-        if (native.use_native(x, w) and w.ndim == 4
-                and x.dtype == torch.bfloat16 and w.shape[-1] == 3
-                and x.shape[1] % 32 == 0 and x.shape[2] % 8 == 0
-                and x.shape[3] % 16 == 0 and x.shape[2] >= 8
-                and x.shape[3] >= 16):
-            return native.require_ext().conv2d_up2(
-                x.contiguous(), w.contiguous())
+        if native.use_native(x, w):
+            if (w.ndim == 4 and x.dtype == torch.bfloat16
+                    and w.shape[-1] == 3 and x.shape[1] % 32 == 0
+                    and x.shape[2] % 8 == 0 and x.shape[3] % 16 == 0
+                    and x.shape[2] >= 8 and x.shape[3] >= 16):
+                return native.require_ext().conv2d_up2(
+                    x.contiguous(), w.contiguous())
+            # small/odd shapes: zero-stuff + our generic conv kernel
+            # (never MIOpen: its naive bf16 conv is ~10x slower)
+            return native.require_ext().conv2d_fwd(
+                _zero_stuff2(x).contiguous(), w.contiguous(), 1, 1)
         return _eager_conv2d(_zero_stuff2(x), w, 1, 1)
 
     @staticmethod

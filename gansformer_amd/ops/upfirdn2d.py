@@ -71,12 +71,45 @@ def _out_size(w, u, d, p0, p1, fw):
     return (w * u + p0 + p1 - fw) // d + 1
 
 
+_sep_cache: dict = {}
+
+
+def _separable8(f):
+    """For a rank-1 4x4 filter, return concat(fy, fx) [8] on f.device;
+    else None. Cached per filter tensor (the resample filters are
+    registered buffers, so the one-time CPU check amortizes away)."""
+    key = (f.data_ptr(), tuple(f.shape), f.device.index)
+    if key in _sep_cache:
+        return _sep_cache[key]
+    res = None
+    if f.shape == (4, 4):
+        fc = f.detach().cpu()
+        r = int(fc.abs().sum(1).argmax())
+        c = int(fc.abs().sum(0).argmax())
+        piv = fc[r, c].item()
+        if piv != 0.0:
+            fy = fc[:, c] / piv
+            fx = fc[r, :]
+            if torch.allclose(torch.outer(fy, fx), fc, atol=1e-7, rtol=1e-5):
+                res = torch.cat([fy, fx]).to(f.device)
+    _sep_cache[key] = res
+    return res
+
+
 class _Upfirdn2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, f, up, dn, pad, gain):
         ctx.params = (up, dn, pad, gain, x.shape)
         ctx.save_for_backward(f)
         if native.use_native(x):
+            if (x.dtype == torch.bfloat16 and up[0] == up[1]
+                    and dn[0] == dn[1]
+                    and (up[0], dn[0]) in ((1, 1), (2, 1), (1, 2))):
+                f8 = _separable8(f)
+                if f8 is not None:
+                    return native.require_ext().upfirdn2d_sep(
+                        x.contiguous(), f8, up[0], dn[0],
+                        pad[2], pad[3], pad[0], pad[1], gain)
             return native.require_ext().upfirdn2d(
                 x.contiguous(), f.contiguous(),
                 up[1], up[0], dn[1], dn[0],

@@ -272,3 +272,24 @@ def test_conv2d_up2(dev, cfg):
     ref = F.conv2d(_zero_stuff2(x.float().cpu()), w.float().cpu(), padding=1)
     assert y.shape == ref.shape
     close(y, ref, torch.bfloat16)
+
+
+@pytest.mark.parametrize("u,d,pads", [
+    (1, 1, (2, 1, 2, 1)), (2, 1, (2, 1, 2, 1)), (1, 2, (1, 1, 1, 1)),
+    (1, 1, (1, 2, 2, 1)),
+])
+def test_upfirdn2d_sep(dev, u, d, pads):
+    """Separable two-pass 4-tap kernel vs eager reference (bf16)."""
+    from gansformer_amd.ops.upfirdn2d import (_eager_upfirdn2d, _separable8,
+                                              setup_filter)
+    torch.manual_seed(7)
+    x = torch.randn(2, 7, 64, 48, device=dev, dtype=torch.bfloat16)
+    f = setup_filter([1, 3, 3, 1], device=dev)
+    f8 = _separable8(f)
+    assert f8 is not None
+    px0, px1, py0, py1 = pads
+    y = _C().upfirdn2d_sep(x, f8, u, d, px0, px1, py0, py1, 1.5)
+    ref = _eager_upfirdn2d(x.float().cpu(), f.cpu(), (u, u), (d, d),
+                           (py0, py1, px0, px1), 1.5)
+    assert y.shape == ref.shape
+    close(y, ref, torch.bfloat16)
