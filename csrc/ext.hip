@@ -49,6 +49,10 @@ void launch_upfirdn2d_sep4_bf16(bf16*, const bf16*, const float*, int, int,
                                 float, hipStream_t);
 bool conv2d_up2_eligible(int I, int O, int H, int W, int kh, int kw,
                          int per_sample);
+bool conv2d_s2_eligible(int I, int O, int H, int W, int OH, int OW, int kh,
+                        int kw, int stride, int pad, int per_sample);
+void launch_conv2d_s2_slab_bf16(bf16*, const bf16*, const bf16*, int B,
+                                int I, int H, int W, int O, hipStream_t);
 void launch_conv2d_up2_slab_bf16(bf16*, const bf16*, const bf16*, int B,
                                  int I, int H, int W, int O, hipStream_t);
 bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
@@ -224,6 +228,12 @@ Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
     gfa::launch_conv2d_fwd_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                      cptr<gfa::bf16>(wr), p.B, p.I, p.H, p.W,
                                      p.O, cur_stream());
+  } else if (gfa::conv2d_s2_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
+                                     p.kw, p.stride, p.pad, p.per_sample)) {
+    auto wr = w.permute({0, 2, 3, 1}).reshape({O, kh * kw, I}).contiguous();
+    gfa::launch_conv2d_s2_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
+                                    cptr<gfa::bf16>(wr), p.B, p.I, p.H, p.W,
+                                    p.O, cur_stream());
   } else {
     gfa::launch_conv2d_fwd_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                 cptr<gfa::bf16>(w), p, cur_stream());

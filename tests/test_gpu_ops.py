@@ -99,6 +99,8 @@ def test_upfirdn2d_asym(dev):
     dict(I=32, O=48, H=16, k=3, stride=1, pad=1),    # slab path (bf16 shared)
     dict(I=64, O=160, H=8, k=3, stride=1, pad=1),    # slab path, O tail > 128
     dict(I=128, O=128, H=32, k=3, stride=1, pad=1),  # slab path, bigger
+    dict(I=32, O=48, H=32, k=3, stride=2, pad=1),    # stride-2 slab (W==32)
+    dict(I=64, O=160, H=64, k=3, stride=2, pad=1),   # stride-2 slab, edges
 ])
 def test_conv2d_fwd(dev, dtype, per_sample, cfg):
     torch.manual_seed(3)
