@@ -194,18 +194,169 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
   }
 }
 
+// ---- stride-2 variant ------------------------------------------------
+// Same tap-major structure; a k-step is 32 OUTPUT pixels of one dY row,
+// whose x window is 66 input columns x 3 rows. Each (c, dr) row splits
+// into two overlapping 40-element half-windows (threads 0..191, one
+// each) so register staging stays at 10 s16x4 per thread; the s-shifted
+// LDS copies subsample the window at stride 2.
+__global__ __launch_bounds__(256, 2)
+void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
+                               const bf16* __restrict__ x,
+                               const bf16* __restrict__ dy,
+                               int B, int I, int H, int W, int O,
+                               int nsplit) {
+  __shared__ u16 slab[2][SLAB_N];
+
+  const int OH = H >> 1, OW = W >> 1;
+  const int nCt = I >> 5;
+  const int bx = blockIdx.x;
+  const int xcd = bx & 7;
+  const int rest = bx >> 3;
+  const int ct = rest % nCt;
+  const int pair = xcd + 8 * (rest / nCt);
+  const int nMt = (O + 127) >> 7;
+  const int m0 = (pair % nMt) << 7;
+  const int split = pair / nMt;
+  const int c0 = ct << 5;
+  if (pair >= nMt * nsplit) return;
+
+  const int OWc = OW >> 5;
+  const long iters = (long)B * OH * OWc;
+  const long span = (iters + nsplit - 1) / nsplit;
+  const long it0 = (long)split * span;
+  const long it1 = min(iters, it0 + span);
+  if (it0 >= it1) return;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wm = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 16;
+  const int ag = lane >> 4;
+
+  int a_row[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    a_row[mi] = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
+
+  // 192 (c, dr, half) tasks; half-window h of 40 elems at 2*col0-4+32h
+  const int s_c = t & 31, s_dr = (t >> 5) % 3, s_h = t >> 5 >= 3;
+  const bool stager = t < 192;
+  s16x4 tk[10];
+
+  auto win_load = [&](long it) {
+    if (!stager) return;
+    const int b = (int)(it / ((long)OH * OWc));
+    const int rem = (int)(it - (long)b * OH * OWc);
+    const int row = rem / OWc;
+    const int col0 = (rem - row * OWc) << 5;
+    const int gy = 2 * row + s_dr - 1;
+    const int lo = 2 * col0 - 4 + 32 * s_h;
+    const bf16* src = x + (((long)b * I + c0 + s_c) * H + gy) * W;
+    if (gy < 0 || gy >= H) {
+#pragma unroll
+      for (int j = 0; j < 10; ++j) tk[j] = s16x4{};
+    } else if (lo >= 0 && lo + 40 <= W) {
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src + lo);
+#pragma unroll
+      for (int j = 0; j < 10; ++j) tk[j] = sp[j];
+    } else {
+      u16* rp = reinterpret_cast<u16*>(&tk[0]);
+#pragma unroll
+      for (int i = 0; i < 40; ++i) {
+        const int gx = lo + i;
+        rp[i] = (gx >= 0 && gx < W) ? __builtin_bit_cast(u16, src[gx])
+                                    : (u16)0;
+      }
+    }
+  };
+
+  // copy s, dst j: x column 2*(col0+j)+s-1 = window[2j+s+3-32h]
+  auto win_write = [&](int sb) {
+    if (!stager) return;
+    const u16* rp = reinterpret_cast<const u16*>(&tk[0]);
+    u16* base = &slab[sb][(s_c * 3 + s_dr) * 3 * SLABX];
+    const int j0 = 16 * s_h;
+#pragma unroll
+    for (int s = 0; s < 3; ++s) {
+      u16* dst = base + s * SLABX;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        dst[j0 + j] = rp[2 * (j0 + j) + s + 3 - 32 * s_h];
+    }
+  };
+
+  f32x4 acc[9][4] = {};
+
+  win_load(it0);
+  win_write(0);
+  __syncthreads();
+
+  for (long it = it0; it < it1; ++it) {
+    const int sb = (int)((it - it0) & 1);
+    if (it + 1 < it1) win_load(it + 1);
+
+    const int b = (int)(it / ((long)OH * OWc));
+    const int rem = (int)(it - (long)b * OH * OWc);
+    const long dybase = ((long)b * O) * OH * OW + rem * 32 + ag * 8;
+    s16x8 af[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      af[mi] = *reinterpret_cast<const s16x8*>(
+          dy + dybase + (long)a_row[mi] * OH * OW);
+
+    const u16* sl = &slab[sb][0];
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      const int r = tap / 3, s = tap - r * 3;
+      const s16x8 bfr = *reinterpret_cast<const s16x8*>(
+          &sl[(((wc + (lane & 15)) * 3 + r) * 3 + s) * SLABX + (ag << 3)]);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        acc[tap][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bfr, acc[tap][mi], 0, 0, 0);
+    }
+    if (it + 1 < it1) win_write(sb ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+    const int c = c0 + wc + (lane & 15);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int o0 = m0 + wm + mi * 16 + ag * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int o = o0 + reg;
+        if (o < O)
+          atomicAdd(&ws[((long)o * 9 + tap) * I + c], acc[tap][mi][reg]);
+      }
+    }
+  }
+}
+
 bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                                 int kh, int kw, int stride, int pad,
                                 int per_sample) {
-  return !per_sample && kh == 3 && kw == 3 && stride == 1 && pad == 1 &&
-         (I & 31) == 0 && I >= 32 && (W & 31) == 0 && W >= 32 &&
-         OH == H && OW == W;
+  if (per_sample || kh != 3 || kw != 3 || pad != 1 || (I & 31) != 0 ||
+      I < 32)
+    return false;
+  if (stride == 1)
+    return (W & 31) == 0 && W >= 32 && OH == H && OW == W;
+  if (stride == 2)
+    return (W & 63) == 0 && W >= 64 && (H & 1) == 0 && OH == H / 2 &&
+           OW == W / 2;
+  return false;
 }
 
-int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O) {
+int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O, int stride) {
   const int nCt = I >> 5;
   const int nMt = (O + 127) >> 7;
-  const long iters = (long)B * H * (W >> 5);
+  const int OH = (stride == 2) ? H / 2 : H;
+  const int OW = (stride == 2) ? W / 2 : W;
+  const long iters = (long)B * OH * (OW >> 5);
   int nsplit = (int)((1024 + (long)nCt * nMt - 1) / ((long)nCt * nMt));
   if (nsplit > iters) nsplit = (int)iters;
   if (nsplit > 64) nsplit = 64;
@@ -215,15 +366,19 @@ int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O) {
 
 void launch_conv2d_wgrad_slab_bf16(float* ws, const bf16* x, const bf16* dy,
                                    int B, int I, int H, int W, int O,
-                                   int nsplit, hipStream_t s) {
+                                   int nsplit, int stride, hipStream_t s) {
   const int nCt = I >> 5;
   const int nMt = (O + 127) >> 7;
   // pair ids run 0..nMt*nsplit-1; grid.x covers xcd-slot * ct * pairHi
   const int npair = nMt * nsplit;
   const int pair_hi = (npair + 7) >> 3;
   dim3 grid((unsigned)(8 * nCt * pair_hi));
-  hipLaunchKernelGGL(conv2d_wgrad_slab_bf16, grid, dim3(256), 0, s, ws, x,
-                     dy, B, I, H, W, O, nsplit);
+  if (stride == 1)
+    hipLaunchKernelGGL(conv2d_wgrad_slab_bf16, grid, dim3(256), 0, s, ws, x,
+                       dy, B, I, H, W, O, nsplit);
+  else
+    hipLaunchKernelGGL(conv2d_wgrad_slab_s2_bf16, grid, dim3(256), 0, s, ws,
+                       x, dy, B, I, H, W, O, nsplit);
 }
 
 }  // namespace gfa

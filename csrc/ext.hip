@@ -58,10 +58,10 @@ void launch_conv2d_up2_slab_bf16(bf16*, const bf16*, const bf16*, int B,
 bool conv2d_wgrad_slab_eligible(int I, int O, int H, int W, int OH, int OW,
                                 int kh, int kw, int stride, int pad,
                                 int per_sample);
-int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O);
+int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O, int stride);
 void launch_conv2d_wgrad_slab_bf16(float*, const bf16*, const bf16*, int B,
                                    int I, int H, int W, int O, int nsplit,
-                                   hipStream_t);
+                                   int stride, hipStream_t);
 void launch_conv2d_fwd_f32(float*, const float*, const float*,
                            const ConvParams&, hipStream_t);
 int conv2d_wgrad_nsplit(const ConvParams&);
@@ -280,11 +280,12 @@ Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
     // tap-major kernel accumulates f32 into [O][9][I]; rearrange after
     auto ws = torch::zeros({O, 9, x.size(1)},
                            x.options().dtype(torch::kFloat32));
-    int ns = gfa::conv2d_wgrad_slab_nsplit(p.B, p.I, p.H, p.W, p.O);
+    int ns = gfa::conv2d_wgrad_slab_nsplit(p.B, p.I, p.H, p.W, p.O,
+                                           p.stride);
     gfa::launch_conv2d_wgrad_slab_bf16(ws.data_ptr<float>(),
                                        cptr<gfa::bf16>(x),
                                        cptr<gfa::bf16>(dy), p.B, p.I, p.H,
-                                       p.W, p.O, ns, cur_stream());
+                                       p.W, p.O, ns, p.stride, cur_stream());
     return ws.reshape({O, 3, 3, x.size(1)})
         .permute({0, 3, 1, 2})
         .contiguous()

@@ -138,16 +138,20 @@ def test_conv2d_wgrad(dev, dtype, per_sample, stride):
     dict(B=2, I=32, O=48, H=32),    # W==32 staging path, O tail
     dict(B=2, I=64, O=160, H=64),   # left/right edge + interior, 2 m-tiles
     dict(B=3, I=96, O=128, H=32),   # 3 channel tiles
+    dict(B=2, I=32, O=64, H=64, stride=2),   # stride-2 (W==64)
+    dict(B=2, I=64, O=160, H=128, stride=2), # stride-2, 2 m-tiles
 ])
 def test_conv2d_wgrad_slab(dev, cfg):
     """Tap-major bf16 wgrad kernel vs eager (slab-eligible shapes)."""
     torch.manual_seed(5)
     B, I, O, H = cfg["B"], cfg["I"], cfg["O"], cfg["H"]
+    st = cfg.get("stride", 1)
+    OH = H // st
     x = torch.randn(B, I, H, H, device=dev, dtype=torch.bfloat16)
-    dy = torch.randn(B, O, H, H, device=dev, dtype=torch.bfloat16)
-    dw = _C().conv2d_wgrad(x, dy, 1, 1, 3, 3, False)
+    dy = torch.randn(B, O, OH, OH, device=dev, dtype=torch.bfloat16)
+    dw = _C().conv2d_wgrad(x, dy, st, 1, 3, 3, False)
     from gansformer_amd.ops.conv2d_grad import _eager_wgrad
-    ref = _eager_wgrad(x.float().cpu(), dy.float().cpu(), 1, 1, 3, 3, False)
+    ref = _eager_wgrad(x.float().cpu(), dy.float().cpu(), st, 1, 3, 3, False)
     assert dw.shape == ref.shape
     close(dw, ref, torch.bfloat16)
 
