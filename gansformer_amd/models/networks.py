@@ -32,6 +32,7 @@ from ..ops import (
     bias_act,
     bipartite_attention,
     conv2d_gradfix,
+    downsample2d,
     minibatch_stddev,
     modulated_conv2d,
     setup_filter,
@@ -103,10 +104,23 @@ class Conv2dLayer(nn.Module):
             x = upsample2d(x, self.resample_filter, up=self.up)
         if self.down > 1:
             fh = self.resample_filter.shape[0]
-            p0 = (fh - self.down + 1) // 2 + self.padding
-            p1 = (fh - self.down) // 2 + self.padding
-            x = upfirdn2d(x, self.resample_filter, padding=(p0, p1, p0, p1))
-            y = conv2d_gradfix(x, w, stride=self.down, padding=0)
+            if self.padding == 0:
+                # 1x1 down-conv: blur-downsample then a plain conv; this
+                # is exactly blur + strided conv (1x1 commutes with the
+                # subsample) and keeps tensor sizes even
+                x = downsample2d(x, self.resample_filter, down=self.down)
+                y = conv2d_gradfix(x, w, stride=1, padding=0)
+            else:
+                # same-size blur with the conv's own padding kept ON the
+                # conv (pads sum to fh-1 so the blur output stays even
+                # and the strided conv takes the stride-2 slab kernel);
+                # interior-identical to baking the pad into the blur
+                p0 = (fh - self.down + 1) // 2
+                p1 = fh - 1 - p0
+                x = upfirdn2d(x, self.resample_filter,
+                              padding=(p0, p1, p0, p1))
+                y = conv2d_gradfix(x, w, stride=self.down,
+                                   padding=self.padding)
         else:
             y = conv2d_gradfix(x, w, stride=1, padding=self.padding)
         b = self.bias.to(y.dtype) if self.bias is not None else None

@@ -71,12 +71,14 @@ def modulated_conv2d(
                       padding=(fh // 2, fh // 2 - 1, fh // 2, fh // 2 - 1),
                       gain=up * up)
     elif down > 1:
-        # blur (same-size, with the downsample pad baked in), then strided conv
+        # same-size blur (pads sum to fh-1, output stays even), strided
+        # conv keeps its own padding -> stride-2 slab kernel eligible;
+        # interior-identical to baking the pad into the blur
         fh = resample_filter.shape[0]
-        p0 = (fh - down + 1) // 2 + padding
-        p1 = (fh - down) // 2 + padding
+        p0 = (fh - down + 1) // 2
+        p1 = fh - 1 - p0
         x = upfirdn2d(x, resample_filter, padding=(p0, p1, p0, p1))
-        y = conv2d_gradfix(x, w, stride=down, padding=0)
+        y = conv2d_gradfix(x, w, stride=down, padding=padding)
     else:
         y = conv2d_gradfix(x, w, stride=1, padding=padding)
 
