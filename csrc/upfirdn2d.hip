@@ -122,9 +122,52 @@ __global__ void upfirdn2d_tiled(T* __restrict__ out, const T* __restrict__ x,
   }
 }
 
+// 1x1-filter up-2 upfirdn == zero-stuff scatter (the id-filter path the
+// strided-conv input gradient uses); pure bandwidth, 8-wide stores.
+template <typename T>
+__global__ void upfirdn2d_zstuff2(T* __restrict__ out,
+                                  const T* __restrict__ x,
+                                  const float* __restrict__ f,
+                                  UfdParams p) {
+  using R = typename Raw<T>::type;
+  const float f0 = f[0] * p.gain;
+  const long nvec = (long)p.B * p.C * p.OH * (p.OW >> 3);
+  for (long idx = global_tid(); idx < nvec; idx += global_stride()) {
+    const int vx = (int)(idx % (p.OW >> 3));
+    const int oy = (int)((idx / (p.OW >> 3)) % p.OH);
+    const long bc = idx / ((long)(p.OW >> 3) * p.OH);
+    R vals[8] = {};
+    const int zy = oy - p.py0;
+    if (zy >= 0 && (zy & 1) == 0 && (zy >> 1) < p.H) {
+      const T* row = x + (bc * p.H + (zy >> 1)) * p.W;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const int zx = vx * 8 + k - p.px0;
+        if (zx >= 0 && (zx & 1) == 0 && (zx >> 1) < p.W)
+          vals[k] = f32_to_raw<R>(to_f32(row[zx >> 1]) * f0);
+      }
+    }
+    if (sizeof(R) == 2) {
+      *reinterpret_cast<s16x8*>(reinterpret_cast<R*>(out) + idx * 8) =
+          *reinterpret_cast<const s16x8*>(&vals[0]);
+    } else {
+#pragma unroll
+      for (int k = 0; k < 8; ++k)
+        reinterpret_cast<R*>(out)[idx * 8 + k] = vals[k];
+    }
+  }
+}
+
 template <typename T>
 void launch_upfirdn2d(T* out, const T* x, const float* f, const UfdParams& p,
                       hipStream_t stream) {
+  if (p.fh == 1 && p.fw == 1 && p.upx == 2 && p.upy == 2 && p.downx == 1 &&
+      p.downy == 1 && (p.OW & 7) == 0) {
+    long nvec = (long)p.B * p.C * p.OH * (p.OW >> 3);
+    hipLaunchKernelGGL((upfirdn2d_zstuff2<T>), stream_grid(nvec), dim3(256),
+                       0, stream, out, x, f, p);
+    return;
+  }
   bool sym = (p.upx == p.upy) && (p.downx == p.downy);
   long ntile_work = 0;
   if (sym && p.fh <= 8 && p.fw <= 8 && p.fh == p.fw) {

@@ -74,9 +74,16 @@ def _conv_input_grad(dy, w, stride, pad, in_hw):
     q1x = W + kw - 1 - Wout * stride - q0x
     if stride == 1 and q0y == q1y and q0x == q1x and q0y >= 0 and q0y == q0x:
         return conv2d_gradfix(dy, wt, stride=1, padding=q0y)
+    # fold one unit of each pad into the conv so the stride-1 conv keeps
+    # pad=1 and stays eligible for the 3x3 slab kernel
+    fold = 1 if (kh == 3 and kw == 3 and min(q0y, q1y, q0x, q1x) >= 1) else 0
     z = upfirdn2d(dy, _id_filter(dy.device), up=stride,
-                  padding=(q0y, q1y, q0x, q1x))
-    return conv2d_gradfix(z, wt, stride=1, padding=0)
+                  padding=(q0y - fold, q1y - fold, q0x - fold, q1x - fold))
+    return conv2d_gradfix(z, wt, stride=1, padding=fold)
+
+
+def _is_1x1(kh, kw, stride, pad):
+    return kh == 1 and kw == 1 and stride == 1 and pad == 0
 
 
 class _Conv2dWgrad(torch.autograd.Function):
@@ -85,6 +92,15 @@ class _Conv2dWgrad(torch.autograd.Function):
         ctx.save_for_backward(x, dy)
         ctx.params = (stride, pad, kh, kw, per_sample)
         if native.use_native(x, dy):
+            if _is_1x1(kh, kw, stride, pad) and not per_sample:
+                # 1x1 wgrad is a plain GEMM: dw = sum_b dY_b X_b^T
+                # (hipBLASLt; the MFMA tile kernels waste M rows on the
+                # skinny tRGB/fromRGB shapes)
+                B, I = x.shape[0], x.shape[1]
+                O = dy.shape[1]
+                dw = torch.matmul(dy.reshape(B, O, -1),
+                                  x.reshape(B, I, -1).transpose(1, 2))
+                return dw.sum(0).reshape(O, I, 1, 1)
             return native.require_ext().conv2d_wgrad(
                 x.contiguous(), dy.contiguous(), stride, pad, kh, kw, per_sample)
         return _eager_wgrad(x, dy, stride, pad, kh, kw, per_sample)
@@ -107,6 +123,12 @@ class _Conv2dFwd(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         ctx.params = (stride, pad)
         if native.use_native(x, w):
+            if w.ndim == 4 and _is_1x1(w.shape[-2], w.shape[-1], stride, pad):
+                # 1x1 conv is a plain GEMM (hipBLASLt)
+                B, I, H, W = x.shape
+                O = w.shape[0]
+                y = torch.matmul(w.reshape(O, I), x.reshape(B, I, H * W))
+                return y.reshape(B, O, H, W)
             return native.require_ext().conv2d_fwd(
                 x.contiguous(), w.contiguous(), stride, pad)
         return _eager_conv2d(x, w, stride, pad)
