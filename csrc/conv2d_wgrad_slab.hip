@@ -261,14 +261,24 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
       const s16x4* sp = reinterpret_cast<const s16x4*>(src + lo);
 #pragma unroll
       for (int j = 0; j < 10; ++j) tk[j] = sp[j];
-    } else {
-      u16* rp = reinterpret_cast<u16*>(&tk[0]);
+    } else if (lo < 0) {
+      // left edge: lo is always -4 (col0==0, h==0); rp[0..3] are x<0
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src);
+      s16x4 tmp[9];
 #pragma unroll
-      for (int i = 0; i < 40; ++i) {
-        const int gx = lo + i;
-        rp[i] = (gx >= 0 && gx < W) ? __builtin_bit_cast(u16, src[gx])
-                                    : (u16)0;
-      }
+      for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
+      tk[0] = s16x4{};
+#pragma unroll
+      for (int j = 0; j < 9; ++j) tk[1 + j] = tmp[j];
+    } else {
+      // right edge: lo - (W-40) is always +4; rp[36..39] are x>=W
+      const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 40);
+      s16x4 tmp[9];
+#pragma unroll
+      for (int j = 0; j < 9; ++j) tmp[j] = sp[1 + j];
+#pragma unroll
+      for (int j = 0; j < 9; ++j) tk[j] = tmp[j];
+      tk[9] = s16x4{};
     }
   };
 
@@ -357,9 +367,11 @@ int conv2d_wgrad_slab_nsplit(int B, int I, int H, int W, int O, int stride) {
   const int OH = (stride == 2) ? H / 2 : H;
   const int OW = (stride == 2) ? W / 2 : W;
   const long iters = (long)B * OH * (OW >> 5);
-  int nsplit = (int)((1024 + (long)nCt * nMt - 1) / ((long)nCt * nMt));
+  // target ~2048 blocks (2 WGs/CU + tail balance); atomics scale with
+  // nsplit so cap at 128
+  int nsplit = (int)((2048 + (long)nCt * nMt - 1) / ((long)nCt * nMt));
   if (nsplit > iters) nsplit = (int)iters;
-  if (nsplit > 64) nsplit = 64;
+  if (nsplit > 128) nsplit = 128;
   if (nsplit < 1) nsplit = 1;
   return nsplit;
 }

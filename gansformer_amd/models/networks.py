@@ -315,7 +315,8 @@ class SynthesisLayer(nn.Module):
                                 device=y.device, dtype=y.dtype)
             else:
                 n = self.noise_const.to(y.dtype).reshape(1, 1, *self.noise_const.shape)
-            y = y + n * self.noise_strength.to(y.dtype)
+            # fused y + n*strength in one pass
+            y = torch.addcmul(y, n, self.noise_strength.to(y.dtype))
         return bias_act(y, self.bias.to(y.dtype), act="lrelu",
                         clamp=self.conv_clamp)
 
