@@ -30,7 +30,7 @@ def main(argv=None):
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=16)
     p.add_argument("--warmup", type=int, default=4)
-    p.add_argument("--batch-gpu", type=int, default=8)
+    p.add_argument("--batch-gpu", type=int, default=32)
     p.add_argument("--res", type=int, default=256)
     p.add_argument("--components-num", type=int, default=16)
     p.add_argument("--transformer", default="duplex")
