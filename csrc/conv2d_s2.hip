@@ -77,38 +77,24 @@ void conv2d_s2_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < 10; ++j) tk[k][j] = sp[j];
       } else if (W == 32) {
-        // single-tile rows: [0,32) with zeros off both ends
+        // single-tile rows: [0,32) direct into tk[1..8], zeros off ends
         const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
-        s16x4 tmp[8];
+        tk[k][0] = s16x4{};
 #pragma unroll
-        for (int j = 0; j < 8; ++j) tmp[j] = sp[j];
-        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-        rp[3] = 0;
-#pragma unroll
-        for (int i = 0; i < 32; ++i) rp[4 + i] = tp[i];
-        rp[36] = 0;
+        for (int j = 0; j < 8; ++j) tk[k][1 + j] = sp[j];
+        tk[k][9] = s16x4{};
       } else if (ix0 == 0) {
+        // left edge: direct loads; tk[0]=0 keeps the rp[3+dx] mapping
         const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
-        s16x4 tmp[9];
+        tk[k][0] = s16x4{};
 #pragma unroll
-        for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
-        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-        rp[3] = 0;
-#pragma unroll
-        for (int i = 0; i < 33; ++i) rp[4 + i] = tp[i];
+        for (int j = 0; j < 9; ++j) tk[k][1 + j] = sp[j];
       } else {
-        // right edge: ix0 + 32 == W; window needs [ix0-1, ix0+33]
+        // right edge (ix0+32 == W): [W-36, W) direct; same rp offset
         const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 36);
-        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
-        s16x4 tmp[9];
 #pragma unroll
-        for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
-        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-#pragma unroll
-        for (int i = 0; i < 33; ++i) rp[3 + i] = tp[3 + i];
-        rp[36] = 0;
+        for (int j = 0; j < 9; ++j) tk[k][j] = sp[j];
+        tk[k][9] = s16x4{};
       }
     }
   };

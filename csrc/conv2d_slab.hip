@@ -98,28 +98,19 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < 6; ++j) tk[k][j] = sp[j];
       } else if (x0 == 0 && W >= 24) {
-        // left edge: aligned loads from 0; slot 3 (x = -1) zeroed in
-        // the write phase never moves — shift source window one slot
+        // left edge: load [0,20) DIRECTLY into tk[1..5]; tk[0]=0 keeps
+        // the rp[3+dx] <-> x[x0-1+dx] mapping (register repacking after
+        // the loads would force a vmcnt(0) wait before the MFMA stream)
         const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
-        s16x4 tmp[5];
+        tk[k][0] = s16x4{};
 #pragma unroll
-        for (int j = 0; j < 5; ++j) tmp[j] = sp[j];
-        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-        rp[3] = 0;  // x = -1
-#pragma unroll
-        for (int dx = 1; dx < 18; ++dx) rp[3 + dx] = tp[dx - 1];
+        for (int j = 0; j < 5; ++j) tk[k][1 + j] = sp[j];
       } else if (x0 + 16 == W && W >= 24) {
-        // right edge: aligned loads from W-24; x = x0-1+dx = (W-24)+7+dx
+        // right edge: load [W-24, W) directly; the write phase switches
+        // to offset 7 (rp[i] <-> x[W-24+i]) and zeroes dx==17 (x==W)
         const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 24);
-        u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
-        s16x4 tmp[6];
 #pragma unroll
-        for (int j = 0; j < 6; ++j) tmp[j] = sp[j];
-        const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-#pragma unroll
-        for (int dx = 0; dx < 17; ++dx) rp[3 + dx] = tp[7 + dx];
-        rp[3 + 17] = 0;  // x = W
+        for (int j = 0; j < 6; ++j) tk[k][j] = sp[j];
       } else {
         u16* rp = reinterpret_cast<u16*>(&tk[k][0]);
 #pragma unroll
@@ -132,6 +123,8 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
     }
   };
 
+  const bool r_edge24 = (x0 + 16 == W) && (x0 >= 4) && (W >= 24);
+
   auto slab_write = [&](int sb) {
 #pragma unroll
     for (int k = 0; k < 2; ++k) {
@@ -140,8 +133,14 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
       const int dy = idx >> 5, c = idx & 31;
       u16* dst = &slab[sb][(dy * 18) * SLAB_PIX + c];
       const u16* rp = reinterpret_cast<const u16*>(&tk[k][0]);
+      if (r_edge24) {
 #pragma unroll
-      for (int dx = 0; dx < 18; ++dx) dst[dx * SLAB_PIX] = rp[3 + dx];
+        for (int dx = 0; dx < 17; ++dx) dst[dx * SLAB_PIX] = rp[7 + dx];
+        dst[17 * SLAB_PIX] = 0;  // x == W
+      } else {
+#pragma unroll
+        for (int dx = 0; dx < 18; ++dx) dst[dx * SLAB_PIX] = rp[3 + dx];
+      }
     }
   };
 

@@ -91,39 +91,25 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
 #pragma unroll
       for (int j = 0; j < 10; ++j) tk[j] = sp[j];
     } else if (W == 32) {
-      // single-chunk rows: load [0, 32), zeros off both ends
+      // single-chunk rows: [0,32) direct into tk[1..8], zeros off ends
       const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-      u16* rp = reinterpret_cast<u16*>(&tk[0]);
-      s16x4 tmp[8];
+      tk[0] = s16x4{};
 #pragma unroll
-      for (int j = 0; j < 8; ++j) tmp[j] = sp[j];
-      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-      rp[3] = 0;
-#pragma unroll
-      for (int i = 0; i < 32; ++i) rp[4 + i] = tp[i];
-      rp[36] = 0;
+      for (int j = 0; j < 8; ++j) tk[1 + j] = sp[j];
+      tk[9] = s16x4{};
     } else if (col0 == 0) {
-      // left edge: loads [0, 36), x=-1 is zero
+      // left edge: direct loads; tk[0]=0 keeps the rp[3+dx] mapping
+      // (register repacking would force vmcnt(0) before the MFMAs)
       const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-      u16* rp = reinterpret_cast<u16*>(&tk[0]);
-      s16x4 tmp[9];
+      tk[0] = s16x4{};
 #pragma unroll
-      for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
-      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-      rp[3] = 0;
-#pragma unroll
-      for (int i = 0; i < 33; ++i) rp[4 + i] = tp[i];
+      for (int j = 0; j < 9; ++j) tk[1 + j] = sp[j];
     } else {
-      // right edge: col0 + 32 == W, loads [W-36, W), x=W is zero
+      // right edge (col0+32 == W): [W-36, W) direct; same rp offset
       const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 36);
-      u16* rp = reinterpret_cast<u16*>(&tk[0]);
-      s16x4 tmp[9];
 #pragma unroll
-      for (int j = 0; j < 9; ++j) tmp[j] = sp[j];
-      const u16* tp = reinterpret_cast<const u16*>(&tmp[0]);
-#pragma unroll
-      for (int i = 0; i < 33; ++i) rp[3 + i] = tp[3 + i];
-      rp[36] = 0;
+      for (int j = 0; j < 9; ++j) tk[j] = sp[j];
+      tk[9] = s16x4{};
     }
   };
 
