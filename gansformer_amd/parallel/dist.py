@@ -25,7 +25,12 @@ def setup_distributed(backend=None, timeout_sec=1800):
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     use_gpu = torch.cuda.is_available()
-    device = torch.device("cuda", local_rank) if use_gpu else torch.device("cpu")
+    if use_gpu:
+        # modulo lets correctness tests run >1 rank per GPU (RCCL
+        # supports it); production runs map one rank per device
+        device = torch.device("cuda", local_rank % torch.cuda.device_count())
+    else:
+        device = torch.device("cpu")
     if use_gpu:
         torch.cuda.set_device(device)
     if world_size > 1 and not dist.is_initialized():
