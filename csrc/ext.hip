@@ -68,6 +68,9 @@ int conv2d_wgrad_nsplit(const ConvParams&);
 template <typename T>
 void launch_conv2d_wgrad(T*, float*, const T*, const T*, const ConvParams&,
                          int, hipStream_t);
+void launch_modnorm_fwd_bf16(bf16*, float*, float*, const bf16*,
+                             const bf16*, const bf16*, long, int, float,
+                             hipStream_t);
 template <typename T>
 void launch_mbstd(float*, const T*, int, int, int, int, int, float,
                   hipStream_t);
@@ -313,6 +316,28 @@ Tensor conv2d_wgrad(Tensor x, Tensor dy, int64_t stride, int64_t pad,
   return ws.to(x.scalar_type());
 }
 
+std::vector<Tensor> modnorm(Tensor x, Tensor gamma, Tensor beta,
+                            double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(gamma.is_contiguous() && beta.is_contiguous());
+  TORCH_CHECK(x.sizes() == gamma.sizes() && x.sizes() == beta.sizes());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              gamma.scalar_type() == torch::kBFloat16 &&
+              beta.scalar_type() == torch::kBFloat16,
+              "modnorm: bf16 only");
+  const long N = x.size(-1);
+  TORCH_CHECK(N % 8 == 0 && N > 0, "modnorm: last dim must be 8-aligned");
+  const long BC = x.numel() / N;
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({BC}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({BC}, x.options().dtype(torch::kFloat32));
+  gfa::launch_modnorm_fwd_bf16(ptr<gfa::bf16>(y), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), cptr<gfa::bf16>(x),
+                               cptr<gfa::bf16>(gamma), cptr<gfa::bf16>(beta),
+                               BC, (int)N, (float)eps, cur_stream());
+  return {y, mean, rstd};
+}
+
 Tensor mbstd(Tensor x, int64_t G, int64_t F, double eps) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
   check_dtype(x, "mbstd.x");
@@ -384,5 +409,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_up2", &conv2d_up2,
         "2x-upsampling 3x3 conv (= conv of zero-stuffed input, pad 1)");
   m.def("mbstd", &mbstd, "minibatch stddev stats [B/G, F]");
+  m.def("modnorm", &modnorm,
+        "fused instance-norm + modulation over the last dim");
   m.def("bipartite_attn", &bipartite_attn, "softmax(QK^T)V");
 }

@@ -34,6 +34,7 @@ from ..ops import (
     conv2d_gradfix,
     downsample2d,
     minibatch_stddev,
+    modnorm,
     modulated_conv2d,
     setup_filter,
     upsample2d,
@@ -68,6 +69,17 @@ class FullyConnected(nn.Module):
             b = (b * self.bias_gain).to(x.dtype)
         y = x.matmul(w.t())
         return _fc_bias_act(y, b, self.act)
+
+
+def _fc_transposed(fc, u):
+    """Linear FC applied to u [B,N,d] with TRANSPOSED output [B,F,N]:
+    (w*gain) @ u^T + bias — a strided batched GEMM, no permute copy."""
+    assert fc.act == "linear"
+    w = (fc.weight * fc.weight_gain).to(u.dtype)
+    y = torch.matmul(w, u.transpose(1, 2))
+    if fc.bias is not None:
+        y = y + (fc.bias * fc.bias_gain).to(y.dtype).reshape(1, -1, 1)
+    return y
 
 
 def _fc_bias_act(y, b, act):
@@ -222,14 +234,18 @@ class BipartiteLayer(nn.Module):
             u = bipartite_attention(qx, ky, vy)  # [B,HW,d] softmax over k
 
         if self.integration in ("mul", "both"):
-            gamma = self.to_gamma(u)  # [B,HW,C]
-            beta = self.to_beta(u)
-            xf = tokens.float()
-            xn = (xf - xf.mean(dim=1, keepdim=True)) \
-                * (xf.var(dim=1, keepdim=True, unbiased=False) + 1e-8).rsqrt()
-            out = xn.to(tokens.dtype) * (1 + gamma) + beta
-        else:
-            out = tokens
+            # gamma/beta computed TRANSPOSED ([B,C,HW], a strided GEMM —
+            # no permute kernels) so the fused instance-norm+modulation
+            # runs in NCHW and writes the layer output layout directly
+            gamma_t = _fc_transposed(self.to_gamma, u)  # [B,C,HW]
+            beta_t = _fc_transposed(self.to_beta, u)
+            xr = x.reshape(B, C, H * W)
+            out_t = modnorm(xr, gamma_t.to(xr.dtype), beta_t.to(xr.dtype))
+            if self.integration == "both":
+                out_t = out_t + torch.tanh(self.x_gate) \
+                    * _fc_transposed(self.to_out, u).to(out_t.dtype)
+            return out_t.reshape(B, C, H, W), y
+        out = tokens
         if self.integration in ("add", "both"):
             out = out + torch.tanh(self.x_gate) * self.to_out(u)
         x = out.transpose(1, 2).reshape(B, C, H, W)

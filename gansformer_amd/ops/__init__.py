@@ -19,4 +19,5 @@ from .upfirdn2d import (  # noqa: F401
 from .conv2d_grad import conv2d_gradfix  # noqa: F401
 from .modulated_conv import modulated_conv2d  # noqa: F401
 from .mbstd import minibatch_stddev  # noqa: F401
+from .modnorm import modnorm  # noqa: F401
 from .bipartite import bipartite_attention  # noqa: F401

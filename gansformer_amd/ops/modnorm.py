@@ -1,0 +1,71 @@
+"""Fused spatial instance-norm + modulation (GANsformer 'mul' integration).
+
+    y[b,c,t] = (x[b,c,t] - mean_t x) * rstd * (1 + gamma[b,c,t]) + beta
+
+The reference computed this as a chain of TF graph ops over the attention
+output (ref src/training/networks.py [R]); eagerly that is
+~10 full-tensor fp32 passes. The HIP kernel does it in two passes per
+(b, c) row in NCHW layout (csrc/modnorm.hip).
+
+Backward RECOMPUTES mean/rstd from x with differentiable torch ops (the
+saved statistics must not enter the graph as constants, or path-length
+double-backward through G would silently drop the d(mu,rstd)/dx terms).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import native
+
+
+def _eager_modnorm(x, gamma, beta, eps):
+    ft = torch.float64 if x.dtype == torch.float64 else torch.float32
+    xf = x.to(ft)
+    m = xf.mean(dim=-1, keepdim=True)
+    v = xf.var(dim=-1, keepdim=True, unbiased=False)
+    xn = (xf - m) * (v + eps).rsqrt()
+    y = xn * (1.0 + gamma.to(ft)) + beta.to(ft)
+    return y.to(x.dtype)
+
+
+class _ModNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        ctx.save_for_backward(x, gamma)
+        ctx.eps = eps
+        if (native.use_native(x, gamma, beta) and x.dtype == torch.bfloat16
+                and x.shape[-1] % 8 == 0):
+            y, _, _ = native.require_ext().modnorm(
+                x.contiguous(), gamma.contiguous(), beta.contiguous(), eps)
+            return y
+        return _eager_modnorm(x, gamma, beta, eps)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma = ctx.saved_tensors
+        eps = ctx.eps
+        ft = torch.float64 if x.dtype == torch.float64 else torch.float32
+        xf = x.to(ft)
+        m = xf.mean(dim=-1, keepdim=True)
+        v = xf.var(dim=-1, keepdim=True, unbiased=False)
+        r = (v + eps).rsqrt()
+        xn = (xf - m) * r
+        dyf = dy.to(ft)
+        dx = dgamma = dbeta = None
+        if ctx.needs_input_grad[2]:
+            dbeta = dy
+        if ctx.needs_input_grad[1]:
+            dgamma = (dyf * xn).to(gamma.dtype)
+        if ctx.needs_input_grad[0]:
+            dxn = dyf * (1.0 + gamma.to(ft))
+            dx = (r * (dxn - dxn.mean(dim=-1, keepdim=True)
+                       - xn * (dxn * xn).mean(dim=-1, keepdim=True)))
+            dx = dx.to(x.dtype)
+        return dx, dgamma, dbeta, None
+
+
+def modnorm(x, gamma, beta, eps=1e-8):
+    """x, gamma, beta: [..., N] (normalized over the last dim)."""
+    assert x.shape == gamma.shape == beta.shape
+    return _ModNorm.apply(x, gamma, beta, float(eps))

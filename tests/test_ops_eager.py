@@ -177,3 +177,28 @@ def test_conv2d_up2_gradcheck():
     w = torch.randn(2, 3, 3, 3, dtype=torch.float64, requires_grad=True) * 0.3
     assert gradcheck(conv2d_up2, (x, w), eps=1e-6, atol=1e-4)
     assert gradgradcheck(conv2d_up2, (x, w), eps=1e-6, atol=1e-4)
+
+
+def test_modnorm_matches_composition():
+    from gansformer_amd.ops.modnorm import modnorm
+    torch.manual_seed(13)
+    x = torch.randn(2, 5, 24, dtype=torch.float64)
+    g = torch.randn(2, 5, 24, dtype=torch.float64) * 0.3
+    b = torch.randn(2, 5, 24, dtype=torch.float64) * 0.2
+    y = modnorm(x, g, b)
+    m = x.mean(-1, keepdim=True)
+    v = x.var(-1, keepdim=True, unbiased=False)
+    ref = (x - m) * (v + 1e-8).rsqrt() * (1 + g) + b
+    assert torch.allclose(y, ref, atol=1e-10)
+
+
+def test_modnorm_gradcheck():
+    from gansformer_amd.ops.modnorm import modnorm
+    torch.manual_seed(14)
+    x = torch.randn(2, 3, 16, dtype=torch.float64, requires_grad=True)
+    g = torch.randn(2, 3, 16, dtype=torch.float64, requires_grad=True) * 0.3
+    b = torch.randn(2, 3, 16, dtype=torch.float64, requires_grad=True) * 0.2
+    g.retain_grad(); b.retain_grad()
+    assert gradcheck(lambda *t: modnorm(*t), (x, g, b), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(lambda *t: modnorm(*t), (x, g, b), eps=1e-6,
+                         atol=1e-4)
