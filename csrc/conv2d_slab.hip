@@ -32,20 +32,31 @@ namespace gfa {
 
 namespace {
 constexpr int SLAB_PIX = 40;           // u16 per slab pixel (32 + 8 pad)
-constexpr int SLAB_N = 10 * 18 * SLAB_PIX;  // one slab buffer, u16
 }  // namespace
 
-__global__ __launch_bounds__(256, 2)
+// TH = output-tile height (8 or 16). TH=16 gives each A fragment 8 B-
+// fragments of reuse instead of 4, halving the strided A-fragment
+// traffic that bounds this kernel (an A-constant experiment ran 2.25x
+// faster), at the cost of acc[4][8] register pressure.
+template <int TH>
+__global__ __launch_bounds__(256, TH == 16 ? 1 : 2)
 void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
                           const bf16* __restrict__ wr,  // [O][9][I]
                           int B, int I, int H, int W, int O) {
-  __shared__ u16 slab[2][SLAB_N];
+  // TH=8: wave = 64(O) x 64(px), acc 4x4. TH=16: wave = 32(O) x 256(px),
+  // acc 2x16 — each A fragment reused 16x and no row is loaded by two
+  // waves, quartering the strided A traffic per output.
+  constexpr int MI = (TH == 16) ? 2 : 4;       // mi frags per wave
+  constexpr int NI = (TH == 16) ? 16 : TH / 2; // ni frags per wave
+  constexpr int SROWS = TH + 2;
+  constexpr int NTASK = SROWS * 32;
+  __shared__ u16 slab[2][SROWS * 18 * SLAB_PIX];
 
   const int tilesX = W >> 4;
   const int b = blockIdx.z;
   const int m0 = blockIdx.y * 128;
   const int ty = blockIdx.x / tilesX, tx = blockIdx.x % tilesX;
-  const int y0 = ty * 8, x0 = tx * 16;
+  const int y0 = ty * TH, x0 = tx * 16;
 
   const bf16* xb = x + (long)b * I * H * W;
   bf16* yb = y + (long)b * O * H * W;
@@ -53,8 +64,8 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
-  const int wm = (wave >> 1) * 64;        // M offset of this wave
-  const int wpy = (wave & 1) * 4;         // tile-row offset of this wave
+  const int wm = (TH == 16) ? wave * 32 : (wave >> 1) * 64;
+  const int wpy = (TH == 16) ? 0 : (wave & 1) * NI;
   const int px = lane & 15;               // B-fragment pixel x (fixed)
   const int ag = lane >> 4;               // K granule 0..3
 
@@ -63,9 +74,9 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   // partial sums whose stores the epilogue skips, so no zero-fill or
   // per-fragment predicate is needed (saves registers). Offsets fit
   // int32 (O*9*I <= 2.4M elements).
-  int a_off[4];
+  int a_off[MI];
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < MI; ++mi) {
     const int o = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
     a_off[mi] = o * 9 * I + ag * 8;
   }
@@ -78,12 +89,13 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   // HBM data before the block's first MFMA and parks the wave (~85%
   // SQ_WAIT_ANY measured). Each thread owns <=2 (dy, c) row tasks of 24
   // elements held as 6 s16x4.
-  const int n_task = (t < 320 - 256) ? 2 : 1;
-  s16x4 tk[2][6];
+  constexpr int KMAX = (NTASK + 255) / 256;
+  const int n_task = (t < NTASK - (KMAX - 1) * 256) ? KMAX : KMAX - 1;
+  s16x4 tk[KMAX][6];
 
   auto slab_load = [&](int c0) {
 #pragma unroll
-    for (int k = 0; k < 2; ++k) {
+    for (int k = 0; k < KMAX; ++k) {
       if (k >= n_task) break;
       const int idx = t + k * 256;
       const int dy = idx >> 5, c = idx & 31;
@@ -127,7 +139,7 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 
   auto slab_write = [&](int sb) {
 #pragma unroll
-    for (int k = 0; k < 2; ++k) {
+    for (int k = 0; k < KMAX; ++k) {
       if (k >= n_task) break;
       const int idx = t + k * 256;
       const int dy = idx >> 5, c = idx & 31;
@@ -144,7 +156,7 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
     }
   };
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[MI][NI] = {};
 
   slab_load(0);
   slab_write(0);
@@ -165,17 +177,17 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
     for (int tap = 0; tap < 9; ++tap) {
       const int r = tap / 3, s = tap - r * 3;
       const int tc = tap * I + c0;
-      s16x8 af[4];
+      s16x8 af[MI];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MI; ++mi)
         af[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + tc);
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < NI; ++ni) {
         const int py = wpy + ni;
         const s16x8 bfr = *reinterpret_cast<const s16x8*>(
             &slab[sb][((py + r) * 18 + px + s) * SLAB_PIX + (ag << 3)]);
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+        for (int mi = 0; mi < MI; ++mi)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bfr, acc[mi][ni], 0, 0, 0);
       }
@@ -186,12 +198,12 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 
   // ---- epilogue: scalar bf16 stores (stride H*W between o rows) ----
 #pragma unroll
-  for (int ni = 0; ni < 4; ++ni) {
+  for (int ni = 0; ni < NI; ++ni) {
     const int gy = y0 + wpy + ni;
     const int gx = x0 + px;
     bf16* yp = yb + (long)gy * W + gx;
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
+    for (int mi = 0; mi < MI; ++mi) {
       const int o0 = m0 + wm + mi * 16 + ag * 4;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
@@ -213,9 +225,15 @@ bool conv2d_slab_eligible(int I, int O, int H, int W, int OH, int OW,
 void launch_conv2d_fwd_slab_bf16(bf16* y, const bf16* x, const bf16* wr,
                                  int B, int I, int H, int W, int O,
                                  hipStream_t s) {
-  dim3 grid((W >> 4) * (H >> 3), ceil_div(O, 128), B);
-  hipLaunchKernelGGL(conv2d_fwd_slab_bf16, grid, dim3(256), 0, s, y, x, wr,
-                     B, I, H, W, O);
+  if ((H & 15) == 0 && H >= 16) {
+    dim3 grid((W >> 4) * (H >> 4), ceil_div(O, 128), B);
+    hipLaunchKernelGGL(conv2d_fwd_slab_bf16<16>, grid, dim3(256), 0, s, y,
+                       x, wr, B, I, H, W, O);
+  } else {
+    dim3 grid((W >> 4) * (H >> 3), ceil_div(O, 128), B);
+    hipLaunchKernelGGL(conv2d_fwd_slab_bf16<8>, grid, dim3(256), 0, s, y, x,
+                       wr, B, I, H, W, O);
+  }
 }
 
 }  // namespace gfa
