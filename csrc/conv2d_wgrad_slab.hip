@@ -69,61 +69,67 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
   for (int mi = 0; mi < 4; ++mi)
     a_row[mi] = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
 
-  // staging: 96 (c, dr) row tasks; threads 0..95 own one each,
-  // holding the 40-elem x window [col0-4, col0+36) as 10 s16x4.
-  const int s_c = t & 31, s_dr = t >> 5;      // dr 0..2 for t < 96
-  const bool stager = t < 96;
-  s16x4 tk[10];
+  // staging: 384 (c, dr, quarter) tasks of 16-element x windows spread
+  // over ALL 256 threads (a 96-full-row split left waves 2-3 idle and
+  // every barrier waiting on wave 0). Task (c, dr, q) covers dst pixels
+  // [8q, 8q+8) of row (c, dr): window [col0-4+8q, +16) as 4 s16x4.
+  const int n_task = (t < 384 - 256) ? 2 : 1;
+  s16x4 tk[2][4];
 
   auto win_load = [&](long it) {
     const int b = (int)(it / ((long)H * OWc));
     const int rem = (int)(it - (long)b * H * OWc);
     const int row = rem / OWc;
     const int col0 = (rem - row * OWc) << 5;
-    const int gy = row + s_dr - 1;
-    const bf16* src = x + (((long)b * I + c0 + s_c) * H + gy) * W;
-    if (!stager) return;
-    if (gy < 0 || gy >= H) {
 #pragma unroll
-      for (int j = 0; j < 10; ++j) tk[j] = s16x4{};
-    } else if (col0 >= 4 && col0 + 36 <= W) {
-      const s16x4* sp = reinterpret_cast<const s16x4*>(src + col0 - 4);
+    for (int k = 0; k < 2; ++k) {
+      if (k >= n_task) break;
+      const int idx = t + k * 256;
+      const int c = idx & 31, rest = idx >> 5;
+      const int dr = rest % 3, q = rest / 3;
+      const int gy = row + dr - 1;
+      const int lo = col0 - 4 + 8 * q;
+      const bf16* src = x + (((long)b * I + c0 + c) * H + gy) * W;
+      if (gy < 0 || gy >= H) {
 #pragma unroll
-      for (int j = 0; j < 10; ++j) tk[j] = sp[j];
-    } else if (W == 32) {
-      // single-chunk rows: [0,32) direct into tk[1..8], zeros off ends
-      const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-      tk[0] = s16x4{};
+        for (int j = 0; j < 4; ++j) tk[k][j] = s16x4{};
+      } else if (lo >= 0 && lo + 16 <= W) {
+        const s16x4* sp = reinterpret_cast<const s16x4*>(src + lo);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) tk[1 + j] = sp[j];
-      tk[9] = s16x4{};
-    } else if (col0 == 0) {
-      // left edge: direct loads; tk[0]=0 keeps the rp[3+dx] mapping
-      // (register repacking would force vmcnt(0) before the MFMAs)
-      const s16x4* sp = reinterpret_cast<const s16x4*>(src);
-      tk[0] = s16x4{};
+        for (int j = 0; j < 4; ++j) tk[k][j] = sp[j];
+      } else if (lo < 0) {
+        // left edge: lo == -4 exactly; x<0 slots are zero
+        const s16x4* sp = reinterpret_cast<const s16x4*>(src);
+        tk[k][0] = s16x4{};
 #pragma unroll
-      for (int j = 0; j < 9; ++j) tk[1 + j] = sp[j];
-    } else {
-      // right edge (col0+32 == W): [W-36, W) direct; same rp offset
-      const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 36);
+        for (int j = 0; j < 3; ++j) tk[k][1 + j] = sp[j];
+      } else {
+        // right edge: lo == W-12 exactly; x>=W slots are zero
+        const s16x4* sp = reinterpret_cast<const s16x4*>(src + W - 16);
 #pragma unroll
-      for (int j = 0; j < 9; ++j) tk[j] = sp[j];
-      tk[9] = s16x4{};
+        for (int j = 0; j < 3; ++j) tk[k][j] = sp[1 + j];
+        tk[k][3] = s16x4{};
+      }
     }
   };
 
-  // expand the register window into the 3 tap-shifted LDS copies:
-  // copy s holds pixels [col0 + s - 1, col0 + s + 31) at 16-B alignment
+  // expand each register window into its slice of the 3 tap-shifted
+  // LDS copies: dst[s][8q + jj] = window[s + 3 + jj]
   auto win_write = [&](int sb) {
-    if (!stager) return;
-    const u16* rp = reinterpret_cast<const u16*>(&tk[0]);
-    u16* base = &slab[sb][(s_c * 3 + s_dr) * 3 * SLABX];
 #pragma unroll
-    for (int s = 0; s < 3; ++s) {
-      u16* dst = base + s * SLABX;
+    for (int k = 0; k < 2; ++k) {
+      if (k >= n_task) break;
+      const int idx = t + k * 256;
+      const int c = idx & 31, rest = idx >> 5;
+      const int dr = rest % 3, q = rest / 3;
+      const u16* rp = reinterpret_cast<const u16*>(&tk[k][0]);
+      u16* base = &slab[sb][(c * 3 + dr) * 3 * SLABX + 8 * q];
 #pragma unroll
-      for (int i = 0; i < 32; ++i) dst[i] = rp[3 + s + i];
+      for (int s = 0; s < 3; ++s) {
+        u16* dst = base + s * SLABX;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) dst[jj] = rp[s + 3 + jj];
+      }
     }
   };
 

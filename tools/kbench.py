@@ -122,11 +122,24 @@ def bench_conv_one(dtype=torch.bfloat16):
     print(f"conv fwd res64 B32: {t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
 
 
+def bench_wgrad_one(dtype=torch.bfloat16):
+    dev = "cuda:0"
+    B, I, O, H, k, s = 32, 512, 512, 64, 3, 1
+    x = torch.randn(B, I, H, H, device=dev, dtype=dtype)
+    dy = torch.randn(B, O, H, H, device=dev, dtype=dtype)
+    t = timeit(lambda: C.conv2d_wgrad(x, dy, s, k // 2, k, k, False),
+               iters=30, warmup=5)
+    fl = 2.0 * B * O * I * k * k * H * H
+    print(f"wgrad res64 B32: {t * 1e3:8.3f} ms  {fl / t / 1e12:7.1f} TF")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     torch.manual_seed(0)
     if which == "conv1":
         bench_conv_one()
+    if which == "wgrad1":
+        bench_wgrad_one()
     if which in ("conv", "all"):
         bench_conv()
     if which in ("wgrad", "all"):
