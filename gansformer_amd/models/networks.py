@@ -210,7 +210,10 @@ class BipartiteLayer(nn.Module):
     def forward(self, x, y):
         """x: [B,C,H,W] image features; y: [B,k,latent_dim] components."""
         B, C, H, W = x.shape
-        tokens = x.reshape(B, C, H * W).transpose(1, 2)  # [B,HW,C]
+        # materialize the token view ONCE: every projection matmul on the
+        # strided transpose view would otherwise copy it again (the
+        # to_q/x_to_k/x_to_v GEMMs were each re-packing [B,HW,C])
+        tokens = x.reshape(B, C, H * W).transpose(1, 2).contiguous()
         pos = self._pos(H, W, x.device, tokens.dtype) if self.use_pos else None
 
         if self.transformer == "duplex":
@@ -322,7 +325,11 @@ class SynthesisLayer(nn.Module):
 
     def forward(self, x, w, noise_mode="random"):
         styles = self.affine(w.to(torch.float32))  # [B, in]
-        y = modulated_conv2d(x, (self.weight * self.weight_gain), styles,
+        # weight_gain is omitted on purpose: demodulation pre-normalizes
+        # the weight by rsqrt(mean w^2), which is scale-invariant in both
+        # value AND gradient, so multiplying by the equalized-LR gain is
+        # an exact no-op here (it still matters for ToRGB / D convs)
+        y = modulated_conv2d(x, self.weight, styles,
                              demodulate=True, up=self.up,
                              resample_filter=self.resample_filter)
         if noise_mode != "none":

@@ -27,12 +27,22 @@ def main():
     for i in range(3):
         tr.step(real, i, i * 32)
     torch.cuda.synchronize()
-    with profile(activities=[ProfilerActivity.CUDA, ProfilerActivity.CPU]) as pr:
+    with profile(activities=[ProfilerActivity.CUDA, ProfilerActivity.CPU],
+                 with_stack=True) as pr:
         for i in range(3, 5):
             tr.step(real, i, i * 32)
         torch.cuda.synchronize()
-    print(pr.key_averages().table(sort_by="cuda_time_total", row_limit=40,
+    print(pr.key_averages().table(sort_by="cuda_time_total", row_limit=25,
                                   max_name_column_width=55))
+    ka = pr.key_averages(group_by_stack_n=6)
+    rows = [e for e in ka if e.key in ("aten::copy_", "aten::clone",
+                                       "aten::contiguous")]
+    rows.sort(key=lambda e: -e.self_device_time_total)
+    for e in rows[:10]:
+        print(f"== {e.key} {e.self_device_time_total / 1e3:.1f} ms "
+              f"x{e.count}")
+        for fr in (e.stack or [])[:6]:
+            print("   ", fr)
 
 if __name__ == "__main__":
     main()
