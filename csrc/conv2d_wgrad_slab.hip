@@ -35,6 +35,7 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
                             const bf16* __restrict__ dy,
                             int B, int I, int H, int W, int O, int nsplit) {
   __shared__ u16 slab[2][SLAB_N];
+  __shared__ u16 ald[2][128 * 40];  // dY tile [row][32px + 8 pad]
 
   // block swizzle: bx = xcd + 8*(ct + nCt*pairHi); pair = xcd + 8*pairHi
   const int nCt = I >> 5;
@@ -63,11 +64,14 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
   const int wc = (wave & 1) * 16;             // C offset of this wave
   const int ag = lane >> 4;                   // pixel granule 0..3
 
-  // A (dY) fragment bases: 4 M rows per wave, this lane's row+granule
-  int a_row[4];
-#pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
-    a_row[mi] = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
+  // dY staging: 512 (row, granule) 16-B chunk tasks over 256 threads
+  // (a direct per-fragment read had 16 lanes hitting 16 strided rows:
+  // a dY-constant experiment ran 1.43x faster)
+  const int dy_lrow = t >> 2;                 // tile rows r and r+64
+  const int dy_row = min(m0 + dy_lrow, O - 1);
+  const int dy_row2 = min(m0 + 64 + dy_lrow, O - 1);
+  const int dy_g = t & 3;                     // 16-B granule 0..3
+  s16x8 dk[2];
 
   // staging: 384 (c, dr, quarter) tasks of 16-element x windows spread
   // over ALL 256 threads (a 96-full-row split left waves 2-3 idle and
@@ -137,21 +141,39 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
 
   win_load(it0);
   win_write(0);
-  __syncthreads();
+
+  auto dy_load = [&](long it) {
+    const int b = (int)(it / ((long)H * OWc));
+    const int rem = (int)(it - (long)b * H * OWc);
+    const long dybase = ((long)b * O) * H * W + (long)rem * 32 + dy_g * 8;
+    dk[0] = *reinterpret_cast<const s16x8*>(dy + dybase +
+                                            (long)dy_row * H * W);
+    dk[1] = *reinterpret_cast<const s16x8*>(dy + dybase +
+                                            (long)dy_row2 * H * W);
+  };
+  auto dy_write = [&](int sb) {
+    *reinterpret_cast<s16x8*>(&ald[sb][dy_lrow * 40 + dy_g * 8]) = dk[0];
+    *reinterpret_cast<s16x8*>(&ald[sb][(64 + dy_lrow) * 40 + dy_g * 8]) =
+        dk[1];
+  };
+
+  dy_load(it0);
+  dy_write(0);
+  __syncthreads();  // note: pairs with the prologue win_write barrier
 
   for (long it = it0; it < it1; ++it) {
     const int sb = (int)((it - it0) & 1);
-    if (it + 1 < it1) win_load(it + 1);
+    if (it + 1 < it1) {
+      win_load(it + 1);
+      dy_load(it + 1);
+    }
 
-    // A fragments for this k-step (32 dY pixels starting at this chunk)
-    const int b = (int)(it / ((long)H * OWc));
-    const int rem = (int)(it - (long)b * H * OWc);
-    const long dybase = ((long)b * O) * H * W + rem * 32 + ag * 8;
+    const u16* al = &ald[sb][0];
     s16x8 af[4];
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
       af[mi] = *reinterpret_cast<const s16x8*>(
-          dy + dybase + (long)a_row[mi] * H * W);
+          &al[(wm + mi * 16 + (lane & 15)) * 40 + (ag << 3)]);
 
     const u16* sl = &slab[sb][0];
 #pragma unroll
@@ -165,7 +187,10 @@ void conv2d_wgrad_slab_bf16(float* __restrict__ ws,  // [O][9][I] f32
         acc[tap][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bfr, acc[tap][mi], 0, 0, 0);
     }
-    if (it + 1 < it1) win_write(sb ^ 1);
+    if (it + 1 < it1) {
+      win_write(sb ^ 1);
+      dy_write(sb ^ 1);
+    }
     __syncthreads();
   }
 
