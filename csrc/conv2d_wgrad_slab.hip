@@ -224,6 +224,7 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
                                int B, int I, int H, int W, int O,
                                int nsplit) {
   __shared__ u16 slab[2][SLAB_N];
+  __shared__ u16 ald[2][128 * 40];  // dY tile, staged like the s1 kernel
 
   const int OH = H >> 1, OW = W >> 1;
   const int nCt = I >> 5;
@@ -252,10 +253,11 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
   const int wc = (wave & 1) * 16;
   const int ag = lane >> 4;
 
-  int a_row[4];
-#pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
-    a_row[mi] = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
+  const int dy_lrow = t >> 2;
+  const int dy_row = min(m0 + dy_lrow, O - 1);
+  const int dy_row2 = min(m0 + 64 + dy_lrow, O - 1);
+  const int dy_g = t & 3;
+  s16x8 dk[2];
 
   // 192 (c, dr, half) tasks; half-window h of 40 elems at 2*col0-4+32h
   const int s_c = t & 31, s_dr = (t >> 5) % 3, s_h = t >> 5 >= 3;
@@ -316,22 +318,40 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
 
   f32x4 acc[9][4] = {};
 
+  auto dy_load = [&](long it) {
+    const int b = (int)(it / ((long)OH * OWc));
+    const int rem = (int)(it - (long)b * OH * OWc);
+    const long dybase = ((long)b * O) * OH * OW + (long)rem * 32 + dy_g * 8;
+    dk[0] = *reinterpret_cast<const s16x8*>(dy + dybase +
+                                            (long)dy_row * OH * OW);
+    dk[1] = *reinterpret_cast<const s16x8*>(dy + dybase +
+                                            (long)dy_row2 * OH * OW);
+  };
+  auto dy_write = [&](int sb) {
+    *reinterpret_cast<s16x8*>(&ald[sb][dy_lrow * 40 + dy_g * 8]) = dk[0];
+    *reinterpret_cast<s16x8*>(&ald[sb][(64 + dy_lrow) * 40 + dy_g * 8]) =
+        dk[1];
+  };
+
   win_load(it0);
   win_write(0);
+  dy_load(it0);
+  dy_write(0);
   __syncthreads();
 
   for (long it = it0; it < it1; ++it) {
     const int sb = (int)((it - it0) & 1);
-    if (it + 1 < it1) win_load(it + 1);
+    if (it + 1 < it1) {
+      win_load(it + 1);
+      dy_load(it + 1);
+    }
 
-    const int b = (int)(it / ((long)OH * OWc));
-    const int rem = (int)(it - (long)b * OH * OWc);
-    const long dybase = ((long)b * O) * OH * OW + rem * 32 + ag * 8;
+    const u16* al = &ald[sb][0];
     s16x8 af[4];
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
       af[mi] = *reinterpret_cast<const s16x8*>(
-          dy + dybase + (long)a_row[mi] * OH * OW);
+          &al[(wm + mi * 16 + (lane & 15)) * 40 + (ag << 3)]);
 
     const u16* sl = &slab[sb][0];
 #pragma unroll
@@ -344,7 +364,10 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
         acc[tap][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bfr, acc[tap][mi], 0, 0, 0);
     }
-    if (it + 1 < it1) win_write(sb ^ 1);
+    if (it + 1 < it1) {
+      win_write(sb ^ 1);
+      dy_write(sb ^ 1);
+    }
     __syncthreads();
   }
 
