@@ -52,15 +52,20 @@ class _BipartiteAttn(torch.autograd.Function):
         q, k, v, = ctx.saved_tensors
         scale = ctx.scale
         # Recompute A with differentiable ops (double-backward-capable).
-        qf, kf, vf = _up(q), _up(k), _up(v)
+        # Logits/softmax stay fp32 (the numerics policy the forward
+        # kernel implements); the surrounding GEMMs run in the input
+        # dtype — hipBLASLt accumulates fp32 internally, and this avoids
+        # materializing fp32 copies of the big [B,HW,*] operands.
+        qf, kf = _up(q), _up(k)
         s = torch.einsum("bqd,bkd->bqk", qf, kf) * scale
         a = torch.softmax(s, dim=-1)
-        do = _up(dout)
-        dv = torch.einsum("bqk,bqe->bke", a, do)
-        da = torch.einsum("bqe,bke->bqk", do, vf)
-        ds = a * (da - (da * a).sum(dim=-1, keepdim=True))
-        dq = torch.einsum("bqk,bkd->bqd", ds, kf) * scale
-        dk = torch.einsum("bqk,bqd->bkd", ds, qf) * scale
+        a_lp = a.to(v.dtype)
+        dv = torch.einsum("bqk,bqe->bke", a_lp, dout)
+        daf = _up(torch.einsum("bqe,bke->bqk", dout, v))
+        ds = a * (daf - (daf * a).sum(dim=-1, keepdim=True))
+        ds_lp = ds.to(q.dtype)
+        dq = torch.einsum("bqk,bkd->bqd", ds_lp, k) * scale
+        dk = torch.einsum("bqk,bqd->bkd", ds_lp, q) * scale
         return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
 
 
