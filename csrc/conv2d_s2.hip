@@ -8,7 +8,7 @@
 // means the 16x8 output tile's window is 33x17 input pixels (+pad), so
 // the slab is single-buffered (49 KB); the next block's loads are still
 // issued before the tap loop and only the LDS write sits between the
-// two barriers.
+// two barriers per channel block.
 //
 // Replaces the cuDNN strided convs of the TF reference.
 #include "common.h"

@@ -20,9 +20,11 @@
 // block's slab is prefetched into the other buffer during the MFMA of
 // the current one.
 //
-// Tile: 128(O) x 128(pixels as 16 wide x 8 high), BC=32, 4 waves, each
-// wave a 64x64 sub-tile = 4x4 fragments of 16x16,
-// v_mfma_f32_16x16x32_bf16, 3 waves/SIMD occupancy.
+// Tiles (template TH): TH=16 -> 128(O) x 256(px as 16x16), each wave a
+// 32x256 sub-tile (2x16 fragments; every A fragment reused 16x, no O row
+// loaded twice — the strided weight-fragment traffic is this kernel's
+// bound). TH=8 -> 128 x 128, wave 64x64 (4x4 fragments), for H==8
+// feature maps. v_mfma_f32_16x16x32_bf16 throughout.
 //
 // Replaces the reference's cuDNN 3x3 convs (TF 1.14).
 #include "common.h"

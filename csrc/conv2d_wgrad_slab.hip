@@ -10,11 +10,12 @@
 //    in LDS into its 9 tap-shifted copies (an im2col image in LDS: the
 //    9x read amplification happens on LDS bandwidth, not HBM), so every
 //    B fragment is an aligned 16-B ds_read.
-//  * dY (the A operand) is read directly from global. Blocks are
-//    swizzled so all I/32 channel-tiles of one (O-tile, K-split) land on
-//    the SAME XCD (blockIdx%8 picks the XCD): they stream the same dY
-//    slice through that XCD's L2, so dY's HBM traffic stays ~1x instead
-//    of x(I/32).
+//  * dY (the A operand) is staged through LDS once per k-step (16-B
+//    chunk tasks over all 256 threads): the per-fragment alternative has
+//    16 lanes hitting 16 strided rows per instruction and parks the
+//    wave. Blocks are additionally swizzled so all I/32 channel-tiles of
+//    one (O-tile, K-split) land on the SAME XCD (blockIdx%8 picks the
+//    XCD) and share the dY slice through that XCD's L2.
 //  * One block computes dw for 128 O x 32 C x all 9 taps; per wave
 //    4(M) x 1(C) x 9(tap) accumulators, 36 MFMA per 32-pixel k-step
 //    against 4 A-fragment loads. f32 atomics accumulate across K-splits.

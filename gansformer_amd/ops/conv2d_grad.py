@@ -1,11 +1,15 @@
 """conv2d with custom, infinitely differentiable autograd.
 
 Replaces the reference's cuDNN convs (TF 1.14 library kernels, SURVEY.md
-K3/L1) with our own MFMA implicit-GEMM HIP kernels on gfx950. Supports
-per-sample weights [B,O,I,kh,kw] natively — this is what makes modulated
-conv a single clean GEMM-shaped op instead of the grouped-conv trick the
-TF lineage used — plus shared weights [O,I,kh,kw], stride 1/2, square
-zero padding.
+K3/L1) with our own MFMA implicit-GEMM HIP kernels on gfx950:
+
+  * 3x3 stride-1/2 shared-weight convs -> tap-major LDS-slab kernels
+    (csrc/conv2d_slab.hip, conv2d_s2.hip, conv2d_wgrad_slab.hip)
+  * 2x-upsampling 3x3 conv -> parity decomposition at input resolution
+    (csrc/conv2d_up2.hip, 1/4 the MACs of upsample-then-conv)
+  * 1x1 convs -> plain hipBLASLt GEMMs
+  * everything else (f32, odd shapes, per-sample weights) -> the generic
+    per-tap implicit-GEMM kernels in csrc/conv2d.hip
 
 Gradient structure (all pieces are themselves autograd Functions or plain
 differentiable torch ops, so R1 / path-length double-backward replays are

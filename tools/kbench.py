@@ -104,8 +104,8 @@ def bench_upfirdn(dtype=torch.bfloat16):
     ]
     for B, Cn, H, up, down, note in shapes:
         x = torch.randn(B, Cn, H, H, device=dev, dtype=dtype)
-        t = timeit(lambda: C.upfirdn2d(x, f, up, up, down, down, 2, 1, 2, 1,
-                                       1.0))
+        from gansformer_amd.ops.upfirdn2d import upfirdn2d as ufd
+        t = timeit(lambda: ufd(x, f, up=up, down=down, padding=(2, 1, 2, 1)))
         nbytes = x.numel() * x.element_size() * (1 + up * up / (down * down))
         print(f"upfirdn  {note:22s}: {t * 1e3:8.3f} ms  "
               f"{nbytes / t / 1e9:7.1f} GB/s")
