@@ -202,3 +202,47 @@ def test_modnorm_gradcheck():
     assert gradcheck(lambda *t: modnorm(*t), (x, g, b), eps=1e-6, atol=1e-4)
     assert gradgradcheck(lambda *t: modnorm(*t), (x, g, b), eps=1e-6,
                          atol=1e-4)
+
+
+def test_modulated_conv_input_scaling_identity():
+    """The input-scaling decomposition used on MI355X must equal the
+    reference's per-sample-weight formulation: conv(x*s, w)*d ==
+    conv(x, w*s*d) (conv is linear in x and w)."""
+    import torch.nn.functional as F
+    torch.manual_seed(21)
+    B, I, O, H = 2, 6, 5, 8
+    x = torch.randn(B, I, H, H, dtype=torch.float64)
+    w = torch.randn(O, I, 3, 3, dtype=torch.float64) * 0.3
+    s = torch.rand(B, I, dtype=torch.float64) + 0.5
+    d = torch.rand(B, O, dtype=torch.float64) + 0.5
+    # decomposition (what modulated_conv2d computes)
+    y1 = F.conv2d(x * s.reshape(B, I, 1, 1), w, padding=1) \
+        * d.reshape(B, O, 1, 1)
+    # reference formulation: grouped conv with per-sample weights
+    wps = w.unsqueeze(0) * s.reshape(B, 1, I, 1, 1) \
+        * d.reshape(B, O, 1, 1, 1)
+    y2 = F.conv2d(x.reshape(1, B * I, H, H), wps.reshape(B * O, I, 3, 3),
+                  padding=1, groups=B).reshape(B, O, H, H)
+    assert torch.allclose(y1, y2, atol=1e-12)
+
+
+def test_up_conv_parity_equals_blur_order():
+    """Parity up-conv + blur-after == blur-upsample + conv (interior):
+    convolutions commute, so the MI355X ordering (conv at input res,
+    1/4 the MACs) matches the reference's upsample-then-conv away from
+    the image border."""
+    import torch.nn.functional as F
+    from gansformer_amd.ops.conv2d_grad import conv2d_up2
+    from gansformer_amd.ops.upfirdn2d import setup_filter, upfirdn2d, upsample2d
+    torch.manual_seed(22)
+    B, I, O, H = 1, 4, 3, 16
+    x = torch.randn(B, I, H, H, dtype=torch.float64)
+    w = torch.randn(O, I, 3, 3, dtype=torch.float64) * 0.3
+    f = setup_filter([1, 3, 3, 1]).to(torch.float64)
+    # MI355X ordering
+    y1 = upfirdn2d(conv2d_up2(x, w), f, padding=(2, 1, 2, 1), gain=4)
+    # reference ordering
+    y2 = F.conv2d(upsample2d(x, f, up=2), w, padding=1)
+    assert y1.shape == y2.shape == (B, O, 2 * H, 2 * H)
+    inner = (slice(None), slice(None), slice(4, -4), slice(4, -4))
+    assert torch.allclose(y1[inner], y2[inner], atol=1e-6)
