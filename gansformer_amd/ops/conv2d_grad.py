@@ -105,6 +105,14 @@ class _Conv2dWgrad(torch.autograd.Function):
                 dw = torch.matmul(dy.reshape(B, O, -1),
                                   x.reshape(B, I, -1).transpose(1, 2))
                 return dw.sum(0).reshape(O, I, 1, 1)
+            if not per_sample and dy.shape[2] * dy.shape[3] <= 256:
+                # small feature maps: im2col + batched GEMM (see fwd)
+                B, I = x.shape[0], x.shape[1]
+                O = dy.shape[1]
+                unf = F.unfold(x, (kh, kw), padding=pad, stride=stride)
+                dw = torch.matmul(dy.reshape(B, O, -1),
+                                  unf.transpose(1, 2))
+                return dw.sum(0).reshape(O, I, kh, kw)
             return native.require_ext().conv2d_wgrad(
                 x.contiguous(), dy.contiguous(), stride, pad, kh, kw, per_sample)
         return _eager_wgrad(x, dy, stride, pad, kh, kw, per_sample)
@@ -121,6 +129,10 @@ class _Conv2dWgrad(torch.autograd.Function):
         return d_x, d_dy, None, None, None, None, None
 
 
+def _out_hw(H, W, kh, kw, stride, pad):
+    return (H + 2 * pad - kh) // stride + 1, (W + 2 * pad - kw) // stride + 1
+
+
 class _Conv2dFwd(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, stride, pad):
@@ -133,6 +145,17 @@ class _Conv2dFwd(torch.autograd.Function):
                 O = w.shape[0]
                 y = torch.matmul(w.reshape(O, I), x.reshape(B, I, H * W))
                 return y.reshape(B, O, H, W)
+            if w.ndim == 4:
+                kh, kw = w.shape[-2], w.shape[-1]
+                OH, OW = _out_hw(x.shape[2], x.shape[3], kh, kw, stride, pad)
+                if OH * OW <= 256:
+                    # small feature maps (res <= 8 layers, mbstd tail):
+                    # the tiled kernels serialize K here; im2col +
+                    # batched hipBLASLt GEMM parallelizes it instead
+                    B, O = x.shape[0], w.shape[0]
+                    unf = F.unfold(x, (kh, kw), padding=pad, stride=stride)
+                    y = torch.matmul(w.reshape(O, -1), unf)
+                    return y.reshape(B, O, OH, OW)
             return native.require_ext().conv2d_fwd(
                 x.contiguous(), w.contiguous(), stride, pad)
         return _eager_conv2d(x, w, stride, pad)
