@@ -170,6 +170,17 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   // issue at the top of the block, the LDS write lands after tap 1 so
   // the register live-range of the staged rows stays short while the
   // global latency still hides behind two taps of MFMA.
+  // TH=16: A fragments are software-pipelined one tap ahead in
+  // registers (8 VGPRs per buffer at MI=2 — affordable where the
+  // narrow tile's MI=4 version spilled); the loads for tap t+1 issue
+  // before tap t's MFMA block and cover the L2/L3 latency.
+  s16x8 afp[MI];
+  if (TH == 16) {
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+      afp[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi]);
+  }
+
   for (int cb = 0; cb < nCB; ++cb) {
     const int c0 = cb << 5;
     const int sb = cb & 1;
@@ -180,9 +191,21 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
       const int r = tap / 3, s = tap - r * 3;
       const int tc = tap * I + c0;
       s16x8 af[MI];
+      if (TH == 16) {
 #pragma unroll
-      for (int mi = 0; mi < MI; ++mi)
-        af[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + tc);
+        for (int mi = 0; mi < MI; ++mi) af[mi] = afp[mi];
+        const bool more = tap < 8 || pre;
+        if (more) {
+          const int ntc = tap < 8 ? tc + I : c0 + 32;
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+            afp[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + ntc);
+        }
+      } else {
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+          af[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + tc);
+      }
 #pragma unroll
       for (int ni = 0; ni < NI; ++ni) {
         const int py = wpy + ni;
