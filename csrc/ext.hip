@@ -71,6 +71,8 @@ void launch_conv2d_wgrad(T*, float*, const T*, const T*, const ConvParams&,
 void launch_modnorm_fwd_bf16(bf16*, float*, float*, const bf16*,
                              const bf16*, const bf16*, long, int, float,
                              hipStream_t);
+void launch_modnorm_bwd_bf16(bf16*, bf16*, const bf16*, const bf16*,
+                             const bf16*, long, int, float, hipStream_t);
 template <typename T>
 void launch_mbstd(float*, const T*, int, int, int, int, int, float,
                   hipStream_t);
@@ -338,6 +340,25 @@ std::vector<Tensor> modnorm(Tensor x, Tensor gamma, Tensor beta,
   return {y, mean, rstd};
 }
 
+std::vector<Tensor> modnorm_bwd(Tensor x, Tensor gamma, Tensor dy,
+                                double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && gamma.is_contiguous() &&
+              dy.is_contiguous());
+  TORCH_CHECK(x.sizes() == gamma.sizes() && x.sizes() == dy.sizes());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "modnorm_bwd: bf16");
+  const long N = x.size(-1);
+  TORCH_CHECK(N % 8 == 0 && N > 0 && N <= 16384,
+              "modnorm_bwd: bad last dim");
+  const long BC = x.numel() / N;
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty_like(gamma);
+  gfa::launch_modnorm_bwd_bf16(ptr<gfa::bf16>(dx), ptr<gfa::bf16>(dgamma),
+                               cptr<gfa::bf16>(x), cptr<gfa::bf16>(gamma),
+                               cptr<gfa::bf16>(dy), BC, (int)N, (float)eps,
+                               cur_stream());
+  return {dx, dgamma};
+}
+
 Tensor mbstd(Tensor x, int64_t G, int64_t F, double eps) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
   check_dtype(x, "mbstd.x");
@@ -411,5 +432,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mbstd", &mbstd, "minibatch stddev stats [B/G, F]");
   m.def("modnorm", &modnorm,
         "fused instance-norm + modulation over the last dim");
+  m.def("modnorm_bwd", &modnorm_bwd,
+        "first-order modnorm backward (dx, dgamma)");
   m.def("bipartite_attn", &bipartite_attn, "softmax(QK^T)V");
 }

@@ -45,6 +45,17 @@ class _ModNorm(torch.autograd.Function):
     def backward(ctx, dy):
         x, gamma = ctx.saved_tensors
         eps = ctx.eps
+        if (not torch.is_grad_enabled() and x.is_cuda
+                and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0
+                and x.shape[-1] <= 16384 and native.have_ext()):
+            # fast fused backward; the differentiable eager composition
+            # below is kept for create_graph replays (path-length reg
+            # differentiates through G's backward graph)
+            dx, dgamma = native.require_ext().modnorm_bwd(
+                x.contiguous(), gamma.contiguous(), dy.contiguous(), eps)
+            db = dy if ctx.needs_input_grad[2] else None
+            return (dx if ctx.needs_input_grad[0] else None,
+                    dgamma if ctx.needs_input_grad[1] else None, db, None)
         ft = torch.float64 if x.dtype == torch.float64 else torch.float32
         xf = x.to(ft)
         m = xf.mean(dim=-1, keepdim=True)

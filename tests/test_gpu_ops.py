@@ -312,3 +312,28 @@ def test_modnorm_kernel(dev):
     ref = _eager_modnorm(x.float().cpu(), g.float().cpu(), b.float().cpu(),
                          1e-8)
     close(y, ref, torch.bfloat16)
+
+
+def test_modnorm_bwd_kernel(dev):
+    """Fused modnorm backward vs the eager composition."""
+    from gansformer_amd.ops.modnorm import modnorm
+    torch.manual_seed(9)
+    shape = (2, 19, 2048)
+    xs = []
+    for use_native_bwd in (False, True):
+        torch.manual_seed(9)
+        x = torch.randn(*shape, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        g = (torch.randn(*shape, device=dev, dtype=torch.bfloat16) * 0.3
+             ).requires_grad_(True)
+        b = (torch.randn(*shape, device=dev, dtype=torch.bfloat16) * 0.2
+             ).requires_grad_(True)
+        y = modnorm(x, g, b)
+        dy = torch.randn_like(y)
+        if use_native_bwd:
+            y.backward(dy)       # grad mode off inside backward -> kernel
+        else:
+            torch.autograd.backward(y, dy, create_graph=True)  # eager path
+        xs.append((x.grad.clone(), g.grad.clone(), b.grad.clone()))
+    for a, b_ in zip(xs[0], xs[1]):
+        close(a, b_, torch.bfloat16)
