@@ -34,6 +34,9 @@ struct AttnParams {
 template <typename T>
 void launch_fba(T*, const T*, const float*, const T*, long, long, int, int,
                 int, float, float, float, hipStream_t);
+void launch_fba_mod(bf16*, const bf16*, const float*, const bf16*,
+                    const float*, float, long, long, int, int, float, float,
+                    float, hipStream_t);
 template <typename T>
 void launch_upfirdn2d(T*, const T*, const float*, const UfdParams&,
                       hipStream_t);
@@ -143,6 +146,39 @@ Tensor fba(Tensor x, Tensor b, Tensor ref, int64_t act, int64_t grad,
         grad == 1 ? cptr<scalar_t>(ref) : nullptr, n, inner, C, (int)act,
         (int)grad, (float)alpha, (float)gain, (float)clamp, cur_stream());
   });
+  return out;
+}
+
+Tensor fba_mod(Tensor x, Tensor d, Tensor noise, Tensor bias,
+               double sigma, int64_t act, double alpha, double gain,
+               double clamp) {
+  // y = clamp(act(x*d[b,c] + noise*sigma + bias[c]) * gain); bf16 NCHW
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fba_mod: bf16 only");
+  const long B = x.size(0), C = x.size(1);
+  const long hw = x.size(2) * x.size(3);
+  TORCH_CHECK(hw % 8 == 0, "fba_mod: H*W must be 8-aligned");
+  TORCH_CHECK(d.is_contiguous() && d.numel() == B * C &&
+              d.scalar_type() == torch::kFloat32, "fba_mod: d must be f32 [B,C]");
+  const bool has_noise = noise.numel() > 0;
+  if (has_noise) {
+    TORCH_CHECK(noise.is_contiguous() && noise.numel() == B * hw &&
+                noise.scalar_type() == torch::kBFloat16,
+                "fba_mod: noise must be bf16 [B,1,H,W]");
+  }
+  const bool has_bias = bias.numel() > 0;
+  Tensor b32;
+  if (has_bias) {
+    b32 = bias.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(b32.numel() == C, "fba_mod: bias size mismatch");
+  }
+  auto out = torch::empty_like(x);
+  gfa::launch_fba_mod(
+      ptr<gfa::bf16>(out), cptr<gfa::bf16>(x), d.data_ptr<float>(),
+      has_noise ? cptr<gfa::bf16>(noise) : nullptr,
+      has_bias ? b32.data_ptr<float>() : nullptr, (float)sigma,
+      x.numel(), hw, (int)C, (int)act, (float)alpha, (float)gain,
+      (float)clamp, cur_stream());
   return out;
 }
 
@@ -422,6 +458,8 @@ Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "gansformer_amd gfx950 HIP kernels";
   m.def("fba", &fba, "fused bias+act (grad=0) / grad variant (grad=1)");
+  m.def("fba_mod", &fba_mod,
+        "fused demod-scale + noise + bias + act (synthesis epilogue)");
   m.def("upfirdn2d", &upfirdn2d, "pad-upsample-FIR-downsample");
   m.def("upfirdn2d_sep", &upfirdn2d_sep,
         "separable 4-tap upfirdn (bf16, two-pass LDS)");

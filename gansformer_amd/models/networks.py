@@ -30,6 +30,7 @@ import torch.nn as nn
 
 from ..ops import (
     bias_act,
+    mod_bias_act,
     bipartite_attention,
     conv2d_gradfix,
     downsample2d,
@@ -329,19 +330,20 @@ class SynthesisLayer(nn.Module):
         # the weight by rsqrt(mean w^2), which is scale-invariant in both
         # value AND gradient, so multiplying by the equalized-LR gain is
         # an exact no-op here (it still matters for ToRGB / D convs)
-        y = modulated_conv2d(x, self.weight, styles,
-                             demodulate=True, up=self.up,
-                             resample_filter=self.resample_filter)
-        if noise_mode != "none":
-            if noise_mode == "random":
-                n = torch.randn(y.shape[0], 1, y.shape[2], y.shape[3],
-                                device=y.device, dtype=y.dtype)
-            else:
-                n = self.noise_const.to(y.dtype).reshape(1, 1, *self.noise_const.shape)
-            # fused y + n*strength in one pass
-            y = torch.addcmul(y, n, self.noise_strength.to(y.dtype))
-        return bias_act(y, self.bias.to(y.dtype), act="lrelu",
-                        clamp=self.conv_clamp)
+        y, dvec = modulated_conv2d(x, self.weight, styles,
+                                   demodulate=True, up=self.up,
+                                   resample_filter=self.resample_filter,
+                                   return_demod=True)
+        n = None
+        if noise_mode == "random":
+            n = torch.randn(y.shape[0], 1, y.shape[2], y.shape[3],
+                            device=y.device, dtype=y.dtype)
+        elif noise_mode == "const":
+            n = self.noise_const.to(y.dtype).reshape(
+                1, 1, *self.noise_const.shape).expand(y.shape[0], -1, -1, -1)
+        # demod scale + noise + bias + lrelu in ONE fused pass
+        return mod_bias_act(y, dvec, noise=n, sigma=self.noise_strength,
+                            b=self.bias, act="lrelu", clamp=self.conv_clamp)
 
 
 class ToRGB(nn.Module):

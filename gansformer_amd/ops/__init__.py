@@ -8,7 +8,7 @@ Dispatch policy (single path, no multi-backend machinery):
     golden numerics reference for kernel tests).
 """
 
-from .fused_act import fused_bias_act, bias_act  # noqa: F401
+from .fused_act import fused_bias_act, bias_act, mod_bias_act  # noqa: F401
 from .upfirdn2d import (  # noqa: F401
     upfirdn2d,
     setup_filter,

@@ -152,3 +152,76 @@ def fused_bias_act(x, b=None, act="lrelu", alpha=None, gain=None, clamp=None):
 
 # Alias matching common naming.
 bias_act = fused_bias_act
+
+
+def _eager_mod_bias_act(x, d, noise, sigma, b, act, alpha, gain, clamp):
+    v = x * d.reshape(d.shape[0], d.shape[1], 1, 1).to(x.dtype)
+    if noise is not None:
+        v = v + noise.to(v.dtype) * sigma.to(v.dtype)
+    return _eager_fwd(v, b, act, alpha, gain, clamp)
+
+
+class _ModBiasAct(torch.autograd.Function):
+    """SynthesisLayer epilogue: y = clamp(act(x*d + noise*sigma + b)*gain)
+    in ONE pass (was demod-scale, noise addcmul, and bias_act — three
+    full-tensor passes after every modulated conv)."""
+
+    @staticmethod
+    def forward(ctx, x, d, noise, sigma, b, act, alpha, gain, clamp):
+        d32 = d if d.dtype == torch.float64 else d.to(torch.float32)
+        if (native.use_native(x) and x.dtype == torch.bfloat16
+                and x.shape[2] * x.shape[3] % 8 == 0
+                and act in ("linear", "lrelu")):
+            code = activation_defs[act][0]
+            empty = torch.empty(0, device=x.device, dtype=x.dtype)
+            y = native.require_ext().fba_mod(
+                x.contiguous(), d32.contiguous(),
+                noise.contiguous().to(x.dtype) if noise is not None else empty,
+                b if b is not None else empty,
+                float(sigma), code, alpha, gain,
+                float("inf") if clamp is None else clamp)
+        else:
+            y = _eager_mod_bias_act(x, d, noise, sigma, b, act, alpha, gain,
+                                    clamp)
+        ctx.save_for_backward(x, d32, y,
+                              noise if noise is not None else x.new_empty(0),
+                              sigma)
+        ctx.params = (act, alpha, gain, clamp, b is not None,
+                      noise is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, d32, y, noise, sigma = ctx.saved_tensors
+        act, alpha, gain, clamp, has_bias, has_noise = ctx.params
+        B, C = x.shape[0], x.shape[1]
+        g = _FusedActGrad.apply(dy, y, act, alpha, gain, clamp)
+        dx = dd = dsigma = db = None
+        if ctx.needs_input_grad[0]:
+            dx = g * d32.reshape(B, C, 1, 1).to(g.dtype)
+        rdt = torch.float64 if x.dtype == torch.float64 else torch.float32
+        if ctx.needs_input_grad[1]:
+            dd = (g * x).sum(dim=[2, 3], dtype=rdt)
+        if has_noise and ctx.needs_input_grad[3]:
+            gc = g.sum(dim=1, keepdim=True, dtype=rdt)
+            dsigma = (gc * noise.to(rdt)).sum().reshape(sigma.shape)
+        if has_bias and ctx.needs_input_grad[4]:
+            db = g.sum(dim=[0, 2, 3], dtype=rdt)
+        return dx, dd, None, dsigma, db, None, None, None, None
+
+
+def mod_bias_act(x, d, noise=None, sigma=None, b=None, act="lrelu",
+                 alpha=None, gain=None, clamp=None):
+    """Fused demodulation-scale + noise + bias + activation.
+
+    x: [B,C,H,W] conv output BEFORE demodulation; d: [B,C] demod factors;
+    noise: [B,1,H,W] or None; sigma: noise strength (0-dim tensor)."""
+    spec = activation_defs[act]
+    if alpha is None:
+        alpha = spec[1]
+    if gain is None:
+        gain = spec[2]
+    if sigma is None:
+        sigma = torch.zeros((), device=x.device)
+    return _ModBiasAct.apply(x, d, noise, sigma, b, act, float(alpha),
+                             float(gain), clamp)

@@ -40,6 +40,8 @@ def modulated_conv2d(
     down=1,
     resample_filter=None,  # 2D FIR tensor (setup_filter) when up/down > 1
     padding=None,          # default: 'same' for the kernel size
+    return_demod=False,    # skip applying d; return (y, d) for a fused
+                           # epilogue (mod_bias_act)
 ):
     B, I, H, W = x.shape
     O, I2, kh, kw = weight.shape
@@ -83,5 +85,9 @@ def modulated_conv2d(
         y = conv2d_gradfix(x, w, stride=1, padding=padding)
 
     if demodulate:
+        if return_demod:
+            return y, d
         y = y * d.reshape(B, O, 1, 1).to(y.dtype)
+    elif return_demod:
+        return y, torch.ones(B, O, device=y.device, dtype=torch.float32)
     return y

@@ -337,3 +337,21 @@ def test_modnorm_bwd_kernel(dev):
         xs.append((x.grad.clone(), g.grad.clone(), b.grad.clone()))
     for a, b_ in zip(xs[0], xs[1]):
         close(a, b_, torch.bfloat16)
+
+
+def test_mod_bias_act_kernel(dev):
+    """Fused epilogue kernel vs eager composition (bf16)."""
+    from gansformer_amd.ops.fused_act import (_eager_mod_bias_act,
+                                              mod_bias_act)
+    torch.manual_seed(10)
+    B, C, H = 3, 37, 16
+    x = torch.randn(B, C, H, H, device=dev, dtype=torch.bfloat16)
+    d = torch.rand(B, C, device=dev) + 0.5
+    n = torch.randn(B, 1, H, H, device=dev, dtype=torch.bfloat16)
+    sig = torch.tensor([0.3], device=dev)
+    b = torch.randn(C, device=dev) * 0.1
+    y = mod_bias_act(x, d, noise=n, sigma=sig, b=b, act="lrelu", clamp=256.0)
+    ref = _eager_mod_bias_act(x.float().cpu(), d.cpu(), n.float().cpu(),
+                              sig.cpu(), b.cpu(), "lrelu", 0.2,
+                              2.0 ** 0.5, 256.0)
+    close(y, ref, torch.bfloat16)

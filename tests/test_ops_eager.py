@@ -246,3 +246,35 @@ def test_up_conv_parity_equals_blur_order():
     assert y1.shape == y2.shape == (B, O, 2 * H, 2 * H)
     inner = (slice(None), slice(None), slice(4, -4), slice(4, -4))
     assert torch.allclose(y1[inner], y2[inner], atol=1e-6)
+
+
+def test_mod_bias_act_matches_composition():
+    from gansformer_amd.ops.fused_act import mod_bias_act
+    torch.manual_seed(23)
+    B, C, H = 2, 5, 6
+    x = torch.randn(B, C, H, H, dtype=torch.float64)
+    d = torch.rand(B, C, dtype=torch.float64) + 0.5
+    n = torch.randn(B, 1, H, H, dtype=torch.float64)
+    sig = torch.tensor([0.3], dtype=torch.float64)
+    b = torch.randn(C, dtype=torch.float64) * 0.1
+    y = mod_bias_act(x, d, noise=n, sigma=sig, b=b, act="lrelu", clamp=10.0)
+    v = x * d.reshape(B, C, 1, 1) + n * sig
+    ref = torch.clamp(
+        torch.nn.functional.leaky_relu(v + b.reshape(1, C, 1, 1), 0.2)
+        * math.sqrt(2.0), -10.0, 10.0)
+    assert torch.allclose(y, ref, atol=1e-10)
+
+
+def test_mod_bias_act_gradcheck():
+    from gansformer_amd.ops.fused_act import mod_bias_act
+    torch.manual_seed(24)
+    B, C, H = 2, 3, 4
+    x = torch.randn(B, C, H, H, dtype=torch.float64, requires_grad=True)
+    d = (torch.rand(B, C, dtype=torch.float64) + 0.5).requires_grad_(True)
+    n = torch.randn(B, 1, H, H, dtype=torch.float64)
+    sig = torch.tensor([0.3], dtype=torch.float64, requires_grad=True)
+    b = (torch.randn(C, dtype=torch.float64) * 0.1).requires_grad_(True)
+    fn = lambda x_, d_, s_, b_: mod_bias_act(x_, d_, noise=n, sigma=s_,
+                                             b=b_, act="lrelu", clamp=10.0)
+    assert gradcheck(fn, (x, d, sig, b), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(fn, (x, d, sig, b), eps=1e-6, atol=1e-4)
