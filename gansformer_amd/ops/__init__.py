@@ -20,4 +20,5 @@ from .conv2d_grad import conv2d_gradfix  # noqa: F401
 from .modulated_conv import modulated_conv2d  # noqa: F401
 from .mbstd import minibatch_stddev  # noqa: F401
 from .modnorm import modnorm  # noqa: F401
+from .linear import linear_nobias, linear_transposed  # noqa: F401
 from .bipartite import bipartite_attention  # noqa: F401

@@ -278,3 +278,21 @@ def test_mod_bias_act_gradcheck():
                                              b=b_, act="lrelu", clamp=10.0)
     assert gradcheck(fn, (x, d, sig, b), eps=1e-6, atol=1e-4)
     assert gradgradcheck(fn, (x, d, sig, b), eps=1e-6, atol=1e-4)
+
+
+def test_linear_splitk_matches_matmul():
+    from gansformer_amd.ops.linear import (_LinearSplitK,
+                                           _LinearTransposedOut)
+    torch.manual_seed(25)
+    x = torch.randn(7, 33, 12, dtype=torch.float64, requires_grad=True)
+    w = torch.randn(5, 12, dtype=torch.float64, requires_grad=True)
+    y = _LinearSplitK.apply(x, w)
+    assert torch.allclose(y, x.matmul(w.t()), atol=1e-12)
+    assert gradcheck(_LinearSplitK.apply, (x, w), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(_LinearSplitK.apply, (x, w), eps=1e-6, atol=1e-4)
+    u = torch.randn(3, 14, 12, dtype=torch.float64, requires_grad=True)
+    yt = _LinearTransposedOut.apply(u, w)
+    assert torch.allclose(yt, torch.matmul(w, u.transpose(1, 2)), atol=1e-12)
+    assert gradcheck(_LinearTransposedOut.apply, (u, w), eps=1e-6, atol=1e-4)
+    assert gradgradcheck(_LinearTransposedOut.apply, (u, w), eps=1e-6,
+                         atol=1e-4)
