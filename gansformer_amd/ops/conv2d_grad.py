@@ -109,7 +109,7 @@ class _Conv2dWgrad(torch.autograd.Function):
                 # small feature maps: im2col + batched GEMM (see fwd)
                 B, I = x.shape[0], x.shape[1]
                 O = dy.shape[1]
-                unf = F.unfold(x, (kh, kw), padding=pad, stride=stride)
+                unf = _unfold_batched(x, kh, kw, pad, stride)
                 dw = torch.matmul(dy.reshape(B, O, -1),
                                   unf.transpose(1, 2))
                 return dw.sum(0).reshape(O, I, kh, kw)
@@ -133,6 +133,17 @@ def _out_hw(H, W, kh, kw, stride, pad):
     return (H + 2 * pad - kh) // stride + 1, (W + 2 * pad - kw) // stride + 1
 
 
+def _unfold_batched(x, kh, kw, pad, stride):
+    """F.unfold with the batch folded into channels: torch's im2col
+    launches one kernel PER SAMPLE, which turns the small-conv GEMM
+    routing into a launch storm; per-channel independence makes
+    unfold([1, B*I, H, W]) identical in one launch."""
+    B, I, H, W = x.shape
+    u = F.unfold(x.reshape(1, B * I, H, W), (kh, kw), padding=pad,
+                 stride=stride)
+    return u.reshape(B, I * kh * kw, -1)
+
+
 class _Conv2dFwd(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, stride, pad):
@@ -153,7 +164,7 @@ class _Conv2dFwd(torch.autograd.Function):
                     # the tiled kernels serialize K here; im2col +
                     # batched hipBLASLt GEMM parallelizes it instead
                     B, O = x.shape[0], w.shape[0]
-                    unf = F.unfold(x, (kh, kw), padding=pad, stride=stride)
+                    unf = _unfold_batched(x, kh, kw, pad, stride)
                     y = torch.matmul(w.reshape(O, -1), unf)
                     return y.reshape(B, O, OH, OW)
             return native.require_ext().conv2d_fwd(
