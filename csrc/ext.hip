@@ -35,8 +35,8 @@ template <typename T>
 void launch_fba(T*, const T*, const float*, const T*, long, long, int, int,
                 int, float, float, float, hipStream_t);
 void launch_fba_mod(bf16*, const bf16*, const float*, const bf16*,
-                    const float*, float, long, long, int, int, float, float,
-                    float, hipStream_t);
+                    const float*, const float*, long, long, int, int, float,
+                    float, float, hipStream_t);
 template <typename T>
 void launch_upfirdn2d(T*, const T*, const float*, const UfdParams&,
                       hipStream_t);
@@ -150,7 +150,7 @@ Tensor fba(Tensor x, Tensor b, Tensor ref, int64_t act, int64_t grad,
 }
 
 Tensor fba_mod(Tensor x, Tensor d, Tensor noise, Tensor bias,
-               double sigma, int64_t act, double alpha, double gain,
+               Tensor sigma, int64_t act, double alpha, double gain,
                double clamp) {
   // y = clamp(act(x*d[b,c] + noise*sigma + bias[c]) * gain); bf16 NCHW
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
@@ -161,10 +161,15 @@ Tensor fba_mod(Tensor x, Tensor d, Tensor noise, Tensor bias,
   TORCH_CHECK(d.is_contiguous() && d.numel() == B * C &&
               d.scalar_type() == torch::kFloat32, "fba_mod: d must be f32 [B,C]");
   const bool has_noise = noise.numel() > 0;
+  Tensor sig32;
   if (has_noise) {
     TORCH_CHECK(noise.is_contiguous() && noise.numel() == B * hw &&
                 noise.scalar_type() == torch::kBFloat16,
                 "fba_mod: noise must be bf16 [B,1,H,W]");
+    // sigma stays on device (an .item() here would sync the stream
+    // once per synthesis layer)
+    sig32 = sigma.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(sig32.is_cuda() && sig32.numel() >= 1);
   }
   const bool has_bias = bias.numel() > 0;
   Tensor b32;
@@ -176,7 +181,8 @@ Tensor fba_mod(Tensor x, Tensor d, Tensor noise, Tensor bias,
   gfa::launch_fba_mod(
       ptr<gfa::bf16>(out), cptr<gfa::bf16>(x), d.data_ptr<float>(),
       has_noise ? cptr<gfa::bf16>(noise) : nullptr,
-      has_bias ? b32.data_ptr<float>() : nullptr, (float)sigma,
+      has_bias ? b32.data_ptr<float>() : nullptr,
+      has_noise ? sig32.data_ptr<float>() : nullptr,
       x.numel(), hw, (int)C, (int)act, (float)alpha, (float)gain,
       (float)clamp, cur_stream());
   return out;

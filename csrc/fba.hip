@@ -147,8 +147,10 @@ __global__ void fba_mod_kernel(bf16* __restrict__ out_,
                                const float* __restrict__ d,   // [B*C]
                                const bf16* __restrict__ noise_,  // [B*HW]
                                const float* __restrict__ bias,   // [C]
-                               float sigma, long n, long hw, int C,
+                               const float* __restrict__ sigma_p,  // [1]
+                               long n, long hw, int C,
                                float alpha, float gain, float clamp) {
+  const float sigma = HAS_NOISE ? sigma_p[0] : 0.f;
   typedef u16 vec_t __attribute__((ext_vector_type(8)));
   u16* out = reinterpret_cast<u16*>(out_);
   const u16* x = reinterpret_cast<const u16*>(x_);
@@ -178,9 +180,10 @@ __global__ void fba_mod_kernel(bf16* __restrict__ out_,
 }
 
 void launch_fba_mod(bf16* out, const bf16* x, const float* d,
-                    const bf16* noise, const float* bias, float sigma,
-                    long n, long hw, int C, int act, float alpha, float gain,
-                    float clamp, hipStream_t stream) {
+                    const bf16* noise, const float* bias,
+                    const float* sigma, long n, long hw, int C, int act,
+                    float alpha, float gain, float clamp,
+                    hipStream_t stream) {
   dim3 grid = stream_grid(n / 8);
 #define GFA_MOD_DISPATCH(ACTC)                                              \
   do {                                                                      \

@@ -178,7 +178,7 @@ class _ModBiasAct(torch.autograd.Function):
                 x.contiguous(), d32.contiguous(),
                 noise.contiguous().to(x.dtype) if noise is not None else empty,
                 b if b is not None else empty,
-                float(sigma), code, alpha, gain,
+                sigma, code, alpha, gain,
                 float("inf") if clamp is None else clamp)
         else:
             y = _eager_mod_bias_act(x, d, noise, sigma, b, act, alpha, gain,

@@ -60,9 +60,9 @@ class GANTrainer:
             fake_logits = D(fake)
             loss_d = d_logistic_loss(real_logits, fake_logits) / rounds
             loss_d.backward()
-            rep("Loss/D", loss_d.item() * rounds)
-            rep("Loss/scores_real", real_logits.mean().item())
-            rep("Loss/scores_fake", fake_logits.mean().item())
+            rep("Loss/D", loss_d.detach() * rounds)
+            rep("Loss/scores_real", real_logits.detach().mean())
+            rep("Loss/scores_fake", fake_logits.detach().mean())
         self.d_red.finalize()
         self.d_opt.step()
 
@@ -75,7 +75,7 @@ class GANTrainer:
                 real_logits = D(real, force_fp32=self.r1_fp32)
                 r1 = r1_penalty(real_logits, real)
                 ((self.gamma * r1 * self.d_reg_interval) / rounds).backward()
-                rep("Loss/r1", r1.item())
+                rep("Loss/r1", r1.detach())
             self.d_red.finalize()
             self.d_opt.step()
 
@@ -88,7 +88,7 @@ class GANTrainer:
             fake_logits = D(fake)
             loss_g = g_nonsaturating_loss(fake_logits) / rounds
             loss_g.backward()
-            rep("Loss/G", loss_g.item() * rounds)
+            rep("Loss/G", loss_g.detach() * rounds)
         self.g_red.finalize()
         self.g_opt.step()
 
@@ -103,7 +103,7 @@ class GANTrainer:
                 fake = G.synthesis(ws)
                 pl = self.pl_reg(fake, ws)
                 ((pl * self.g_reg_interval) / rounds).backward()
-                rep("Loss/pl", pl.item())
+                rep("Loss/pl", pl.detach())
             self.g_red.finalize()
             self.g_opt.step()
 
