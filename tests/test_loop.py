@@ -71,3 +71,15 @@ def test_training_loop_metrics(tmp_path):
     with open(tmp_path / "metric-fid1k.txt") as f:
         line = f.read().strip()
     assert "kimg" in line and float(line.split()[-1]) >= 0
+
+
+def test_training_loop_grad_accumulation(tmp_path):
+    """rounds > 1: global batch = 2x the per-device batch, accumulated
+    over two rounds per phase."""
+    out = training_loop(**cfg(tmp_path, batch_gpu=2, batch_size=4,
+                              total_kimg=0.016))
+    assert out["cur_nimg"] >= 16
+    with open(tmp_path / "metrics.jsonl") as f:
+        rows = [json.loads(l) for l in f if l.strip()]
+    assert rows and all(
+        torch.isfinite(torch.tensor(r.get("Loss/D", 0.0))) for r in rows)
