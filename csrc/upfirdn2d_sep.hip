@@ -116,8 +116,8 @@ void upfirdn2d_sep4(T* __restrict__ out, const T* __restrict__ x,
         vals[j] = f32_to_raw<typename Raw<T>::type>(acc * gain);
       }
       const int ox = ox0 + cx8;
-      // tiles are full except at the right/bottom edge of odd sizes
-      if (ox + 8 <= OW) {
+      // vector store needs the row base 16-B aligned too (OW % 8)
+      if (ox + 8 <= OW && (OW & 7) == 0) {
         *reinterpret_cast<s16x8*>(
             reinterpret_cast<typename Raw<T>::type*>(out) +
             bc * (long)OH * OW + (long)oy * OW + ox) =
