@@ -132,6 +132,7 @@ def main(argv=None):
         desc = args.desc or _default_desc(args)
         from gansformer_amd import rundir
         run_dir = rundir.create_run_dir(args.result_dir, desc)
+        rundir.save_submit_config(run_dir, dict(vars(args)))
         mp.spawn(_spawn_main, args=(args.num_gpus, loop_args, run_dir),
                  nprocs=args.num_gpus, join=True)
         return
