@@ -106,3 +106,19 @@ def test_num_ws_indexing():
     # res 4..64 -> 5 blocks; first block 2 layers (conv1+torgb), rest 3
     assert G.synthesis.num_ws == 2 + 4 * 3
     assert G.mapping.num_ws == G.synthesis.num_ws
+
+
+@pytest.mark.parametrize("integration", ["mul", "add", "both"])
+def test_generator_integration_variants(integration):
+    """All three attention-integration modes forward+backward on CPU."""
+    torch.manual_seed(0)
+    G = Generator(z_dim=32, w_dim=32, img_resolution=32, num_components=4,
+                  transformer="duplex", integration=integration,
+                  channel_base=2048, channel_max=64, bf16_res_count=0,
+                  mapping_layers=2)
+    z = G.sample_z(2)
+    img = G(z)
+    assert img.shape == (2, 3, 32, 32) and torch.isfinite(img).all()
+    img.square().mean().backward()
+    assert all(p.grad is None or torch.isfinite(p.grad).all()
+               for p in G.parameters())
