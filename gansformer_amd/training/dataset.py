@@ -93,15 +93,47 @@ class ShardedNpyDataset(Dataset):
         return torch.from_numpy(img), torch.from_numpy(label)
 
 
+class MirroredDataset(Dataset):
+    """x-flip augmentation, reference style [R]: the dataset is doubled
+    and indices past the base length return the horizontally flipped
+    image (deterministic per index, so resume stays reproducible)."""
+
+    def __init__(self, base):
+        self.base = base
+
+    def __len__(self):
+        return 2 * len(self.base)
+
+    @property
+    def resolution(self):
+        return self.base.resolution
+
+    @property
+    def image_shape(self):
+        return self.base.image_shape
+
+    def __getitem__(self, idx):
+        n = len(self.base)
+        img, label = self.base[idx % n]
+        if idx >= n:
+            img = torch.flip(img, dims=[-1])
+        return img, label
+
+
 def load_dataset(dataset=None, data_dir=None, resolution=256, channels=3,
-                 synthetic_size=50000, seed=0):
+                 synthetic_size=50000, seed=0, mirror_augment=False):
     """Resolve a dataset spec: a prepared dir if it exists, else synthetic."""
+    ds = None
     if data_dir and dataset:
         path = os.path.join(data_dir, dataset)
         if os.path.isdir(path):
-            return ShardedNpyDataset(path, resolution=resolution)
-    return SyntheticDataset(resolution=resolution, channels=channels,
-                            size=synthetic_size, seed=seed)
+            ds = ShardedNpyDataset(path, resolution=resolution)
+    if ds is None:
+        ds = SyntheticDataset(resolution=resolution, channels=channels,
+                              size=synthetic_size, seed=seed)
+    if mirror_augment:
+        ds = MirroredDataset(ds)
+    return ds
 
 
 def make_loader(ds, batch_size, rank=0, world_size=1, num_workers=2, seed=0):

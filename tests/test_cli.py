@@ -83,3 +83,16 @@ def test_bench_cli_cpu(tmp_path):
     assert out["steps"] == 2
     assert out["scaling"] == "weak"
     assert out["data"] == "synthetic"
+
+
+def test_mirror_augment_dataset():
+    from gansformer_amd.training.dataset import (MirroredDataset,
+                                                 SyntheticDataset)
+    import torch
+    base = SyntheticDataset(resolution=16, size=8, seed=1)
+    ds = MirroredDataset(base)
+    assert len(ds) == 16 and ds.resolution == 16
+    img, _ = ds[3]
+    img_f, _ = ds[3 + 8]
+    assert torch.equal(img_f, torch.flip(img, dims=[-1]))
+    assert not torch.equal(img_f, img)

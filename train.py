@@ -25,7 +25,7 @@ def build_configs(args):
     dataset_args = EasyDict(
         dataset=args.dataset, data_dir=args.data_dir,
         resolution=args.res, synthetic_size=args.synthetic_size,
-        seed=args.seed)
+        seed=args.seed, mirror_augment=args.mirror_augment)
     if args.inception_path:
         dataset_args.inception_path = args.inception_path
     G_args = EasyDict(
