@@ -68,3 +68,46 @@ def test_conv2d_up2_fuzz(b, i, o, h, w):
     y = conv2d_up2(x, wt)
     ref = F.conv2d(_zero_stuff2(x), wt, padding=1)
     assert torch.allclose(y, ref, atol=1e-10)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    b=st.integers(1, 3), c=st.integers(1, 6), n=st.integers(8, 64),
+)
+def test_modnorm_fuzz(b, c, n):
+    from gansformer_amd.ops.modnorm import modnorm
+    torch.manual_seed(3)
+    x = torch.randn(b, c, n, dtype=torch.float64)
+    g = torch.randn(b, c, n, dtype=torch.float64) * 0.3
+    bt = torch.randn(b, c, n, dtype=torch.float64) * 0.2
+    y = modnorm(x, g, bt)
+    m = x.mean(-1, keepdim=True)
+    v = x.var(-1, keepdim=True, unbiased=False)
+    ref = (x - m) * (v + 1e-8).rsqrt() * (1 + g) + bt
+    assert torch.allclose(y, ref, atol=1e-9)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    b=st.integers(1, 3), c=st.integers(1, 5), h=st.integers(2, 9),
+    use_noise=st.booleans(), use_bias=st.booleans(),
+)
+def test_mod_bias_act_fuzz(b, c, h, use_noise, use_bias):
+    import math as _m
+    from gansformer_amd.ops.fused_act import mod_bias_act
+    torch.manual_seed(4)
+    x = torch.randn(b, c, h, h, dtype=torch.float64)
+    d = torch.rand(b, c, dtype=torch.float64) + 0.5
+    n = torch.randn(b, 1, h, h, dtype=torch.float64) if use_noise else None
+    sig = torch.tensor([0.4], dtype=torch.float64)
+    bias = (torch.randn(c, dtype=torch.float64) * 0.1) if use_bias else None
+    y = mod_bias_act(x, d, noise=n, sigma=sig, b=bias, act="lrelu",
+                     clamp=8.0)
+    v = x * d.reshape(b, c, 1, 1)
+    if use_noise:
+        v = v + n * sig
+    if use_bias:
+        v = v + bias.reshape(1, c, 1, 1)
+    ref = torch.clamp(torch.nn.functional.leaky_relu(v, 0.2)
+                      * _m.sqrt(2.0), -8.0, 8.0)
+    assert torch.allclose(y, ref, atol=1e-10)
