@@ -120,9 +120,13 @@ def collect_features(batch_fn, extractor, num_images, batch_size, device,
         done += n
     feats = torch.cat(feats)[:per_rank]
     if world_size > 1 and tdist.is_initialized():
-        gathered = [torch.zeros_like(feats) for _ in range(world_size)]
-        tdist.all_gather(gathered, feats)
-        feats = torch.cat(gathered)
+        # NCCL/RCCL collectives require device tensors; gloo wants CPU.
+        backend = tdist.get_backend()
+        on_dev = str(backend).lower() in ("nccl", "rccl")
+        src = feats.to(device) if on_dev else feats
+        gathered = [torch.zeros_like(src) for _ in range(world_size)]
+        tdist.all_gather(gathered, src)
+        feats = torch.cat([g.cpu() for g in gathered])
     return feats[:num_images].numpy()
 
 

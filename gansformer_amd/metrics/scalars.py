@@ -28,17 +28,26 @@ class ScalarLogger:
                     self.tb = None
 
     def report(self, name, value):
-        """Accumulate one observation of a scalar."""
-        try:
-            value = float(value)
-        except (TypeError, ValueError):
-            return
-        s = self.accum.setdefault(name, [0.0, 0])
-        s[0] += value
-        s[1] += 1
+        """Accumulate one observation of a scalar.
+
+        Tensors are stored as-is and only converted to floats in
+        flush(), so reporting a GPU tensor costs no host-device sync on
+        the hot path (one sync per tick instead of one per report).
+        """
+        self.accum.setdefault(name, []).append(value)
 
     def means(self):
-        return {k: v[0] / max(v[1], 1) for k, v in self.accum.items()}
+        out = {}
+        for k, vals in self.accum.items():
+            total, n = 0.0, 0
+            for v in vals:
+                try:
+                    total += float(v)
+                    n += 1
+                except (TypeError, ValueError):
+                    pass
+            out[k] = total / max(n, 1)
+        return out
 
     def flush(self, step, **extra):
         """Write the averaged scalars for this tick and reset."""

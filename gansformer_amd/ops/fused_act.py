@@ -223,5 +223,11 @@ def mod_bias_act(x, d, noise=None, sigma=None, b=None, act="lrelu",
         gain = spec[2]
     if sigma is None:
         sigma = torch.zeros((), device=x.device)
+    if noise is not None and noise.requires_grad:
+        # _ModBiasAct.backward returns no grad for the noise input; a
+        # differentiable noise tensor would get its gradient silently
+        # dropped, so refuse it loudly.
+        raise ValueError("mod_bias_act: noise must not require grad "
+                         "(its gradient is not computed)")
     return _ModBiasAct.apply(x, d, noise, sigma, b, act, float(alpha),
                              float(gain), clamp)

@@ -71,28 +71,39 @@ def _out_size(w, u, d, p0, p1, fw):
     return (w * u + p0 + p1 - fw) // d + 1
 
 
-_sep_cache: dict = {}
+import weakref
+
+# Keyed on a weakref to the filter TENSOR OBJECT (the resample filters
+# are long-lived registered buffers, so hits are by identity with no
+# host copy). A data_ptr key would alias a freed tensor's reused
+# address and silently return the wrong factorization; a value key
+# would cost a device sync per call. Entries die with their tensor.
+_sep_cache: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
 
 
 def _separable8(f):
     """For a rank-1 4x4 filter, return concat(fy, fx) [8] on f.device;
-    else None. Cached per filter tensor (the resample filters are
-    registered buffers, so the one-time CPU check amortizes away)."""
-    key = (f.data_ptr(), tuple(f.shape), f.device.index)
-    if key in _sep_cache:
-        return _sep_cache[key]
+    else None."""
+    if f.shape != (4, 4):
+        return None
+    try:
+        return _sep_cache[f]
+    except KeyError:
+        pass
     res = None
-    if f.shape == (4, 4):
-        fc = f.detach().cpu()
-        r = int(fc.abs().sum(1).argmax())
-        c = int(fc.abs().sum(0).argmax())
-        piv = fc[r, c].item()
-        if piv != 0.0:
-            fy = fc[:, c] / piv
-            fx = fc[r, :]
-            if torch.allclose(torch.outer(fy, fx), fc, atol=1e-7, rtol=1e-5):
-                res = torch.cat([fy, fx]).to(f.device)
-    _sep_cache[key] = res
+    fc = f.detach().cpu()
+    r = int(fc.abs().sum(1).argmax())
+    c = int(fc.abs().sum(0).argmax())
+    piv = fc[r, c].item()
+    if piv != 0.0:
+        fy = fc[:, c] / piv
+        fx = fc[r, :]
+        if torch.allclose(torch.outer(fy, fx), fc, atol=1e-7, rtol=1e-5):
+            res = torch.cat([fy, fx]).to(f.device)
+    try:
+        _sep_cache[f] = res
+    except TypeError:  # tensor subclass that refuses weakrefs
+        pass
     return res
 
 

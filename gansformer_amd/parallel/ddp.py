@@ -29,14 +29,14 @@ def broadcast_params(module: torch.nn.Module, src: int = 0):
 
 
 class _Bucket:
-    __slots__ = ("params", "numel", "buffer", "work", "ready", "launched")
+    __slots__ = ("params", "numel", "buffer", "work", "counts", "launched")
 
     def __init__(self):
         self.params = []
         self.numel = 0
         self.buffer = None
         self.work = None
-        self.ready = set()
+        self.counts = {}
         self.launched = False
 
 
@@ -71,8 +71,17 @@ class GradReducer:
         if not self.active:
             return
         b = self.param_bucket[p]
-        b.ready.add(p)
-        if len(b.ready) == len(b.params):
+        # With gradient accumulation (rounds > 1), every backward pass
+        # fires this hook once per participating param. Launching the
+        # all-reduce before the LAST accumulation pass would ship partial
+        # gradients and then overwrite the fully-accumulated p.grad in
+        # finalize(). So a param only counts as ready once it has fired
+        # `rounds` times; params absent from some round's graph simply
+        # never complete the bucket and are flushed in finalize(), where
+        # p.grad holds the full accumulated value.
+        b.counts[p] = b.counts.get(p, 0) + 1
+        if (not b.launched
+                and all(b.counts.get(q, 0) >= self.rounds for q in b.params)):
             self._launch(b)
 
     def _launch(self, b: _Bucket):
@@ -93,15 +102,22 @@ class GradReducer:
         b.work = dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
                                  group=self.group, async_op=True)
 
-    def prepare(self):
-        """Call before backward on a step whose grads should be reduced."""
+    def prepare(self, rounds: int = 1):
+        """Call before backward on a step whose grads should be reduced.
+
+        `rounds` = number of gradient-accumulation backward passes that
+        will run before finalize(); buckets launch eagerly only on the
+        final pass (see _hook).
+        """
         self.active = True
+        self.rounds = max(int(rounds), 1)
         for b in self.buckets:
-            b.ready.clear()
+            b.counts = {}
             b.launched = False
             b.work = None
 
     active = False
+    rounds = 1
 
     def finalize(self):
         """Flush partial buckets, wait for comms, write averaged grads back."""

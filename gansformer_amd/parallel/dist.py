@@ -26,9 +26,21 @@ def setup_distributed(backend=None, timeout_sec=1800):
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     use_gpu = torch.cuda.is_available()
     if use_gpu:
-        # modulo lets correctness tests run >1 rank per GPU (RCCL
-        # supports it); production runs map one rank per device
-        device = torch.device("cuda", local_rank % torch.cuda.device_count())
+        ndev = torch.cuda.device_count()
+        if local_rank >= ndev:
+            # RCCL does NOT allow two ranks on one device (observed:
+            # NCCL "invalid usage" for 2-ranks-1-GPU). Oversubscription
+            # is opt-in for gloo-on-GPU tests only; a misconfigured
+            # production launch must fail loudly here.
+            if os.environ.get("GANSFORMER_OVERSUBSCRIBE_GPU") == "1":
+                device = torch.device("cuda", local_rank % ndev)
+            else:
+                raise RuntimeError(
+                    f"LOCAL_RANK={local_rank} but only {ndev} visible GPU(s);"
+                    " one rank per GPU is required (set"
+                    " GANSFORMER_OVERSUBSCRIBE_GPU=1 only for gloo tests)")
+        else:
+            device = torch.device("cuda", local_rank)
     else:
         device = torch.device("cpu")
     if use_gpu:

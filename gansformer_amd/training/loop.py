@@ -236,6 +236,7 @@ def training_loop(
 
         want_stop = stop_flag["stop"] or ctx.should_stop()
         if cur_nimg >= next_snap or cur_nimg >= total_kimg * 1000 or want_stop:
+            trainer.sync_ranks()
             if is_main and run_dir is not None:
                 extra_state = dict(trainer.state_dict(),
                                    cur_nimg=cur_nimg, step=step)
@@ -246,6 +247,7 @@ def training_loop(
 
         if metrics and (cur_nimg >= next_metric
                         or cur_nimg >= total_kimg * 1000):
+            trainer.sync_ranks()  # all ranks run Gs for FID: keep it identical
             results = run_metrics(cur_nimg)
             if is_main and results:
                 print("metrics: " + json.dumps(results), flush=True)

@@ -50,7 +50,7 @@ class GANTrainer:
 
         # ---------------- D main ----------------
         self.d_opt.zero_grad(set_to_none=True)
-        self.d_red.prepare()
+        self.d_red.prepare(rounds)
         for _ in range(rounds):
             real = next_real_fn()
             z = G.sample_z(self.batch_gpu, device=self.device)
@@ -69,7 +69,7 @@ class GANTrainer:
         # ---------------- D reg (lazy R1) ----------------
         if do_d_reg:
             self.d_opt.zero_grad(set_to_none=True)
-            self.d_red.prepare()
+            self.d_red.prepare(rounds)
             for _ in range(rounds):
                 real = next_real_fn().requires_grad_(True)
                 real_logits = D(real, force_fp32=self.r1_fp32)
@@ -81,7 +81,7 @@ class GANTrainer:
 
         # ---------------- G main ----------------
         self.g_opt.zero_grad(set_to_none=True)
-        self.g_red.prepare()
+        self.g_red.prepare(rounds)
         for _ in range(rounds):
             z = G.sample_z(self.batch_gpu, device=self.device)
             fake = G(z, style_mixing=True, update_w_avg=True)
@@ -95,7 +95,7 @@ class GANTrainer:
         # ---------------- G reg (lazy path length) ----------------
         if do_g_reg:
             self.g_opt.zero_grad(set_to_none=True)
-            self.g_red.prepare()
+            self.g_red.prepare(rounds)
             pl_batch = max(self.batch_gpu // 2, 1)
             for _ in range(rounds):
                 z = G.sample_z(pl_batch, device=self.device)
@@ -114,6 +114,35 @@ class GANTrainer:
                            max(cur_nimg - start_nimg, 1) * self.ema_rampup)
         beta = 0.5 ** (self.batch_size / max(ema_nimg, 1e-8))
         ema_update(self.Gs, G, beta)
+
+    @torch.no_grad()
+    def sync_ranks(self):
+        """Reconcile per-rank-divergent training state across DP ranks.
+
+        Parameters stay in lockstep through the grad all-reduce, but
+        `MappingNetwork.w_avg` and `PathLengthRegularizer.pl_mean` update
+        from rank-local batches and drift apart; `Gs` inherits the drift
+        via its buffers. All-reduce them to the cross-rank mean so that a
+        snapshot saved by rank 0 matches every rank's live state (called
+        at snapshot/metric cadence by the training loop).
+        """
+        import torch.distributed as dist
+        if not dist.is_initialized() or dist.get_world_size() == 1:
+            return
+        inv = 1.0 / dist.get_world_size()
+        bufs = []
+        wa = getattr(getattr(self.G, "mapping", None), "w_avg", None)
+        if wa is not None:
+            bufs.append(wa)
+        if self.pl_reg.pl_mean is not None:
+            bufs.append(self.pl_reg.pl_mean)
+        for t in bufs:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            t.mul_(inv)
+        # propagate the reconciled w_avg into the EMA copy
+        gwa = getattr(getattr(self.Gs, "mapping", None), "w_avg", None)
+        if wa is not None and gwa is not None:
+            gwa.copy_(wa)
 
     def state_dict(self):
         return dict(g_opt=self.g_opt.state_dict(),

@@ -92,3 +92,220 @@ def test_grad_reducer_two_ranks():
         g0, gl = results[rank]
         assert torch.allclose(g0, exp0, atol=1e-6), f"rank {rank} main grads"
         assert torch.allclose(gl, expl, atol=1e-6), f"rank {rank} lazy grads"
+
+
+# ---------------------------------------------------------------------------
+# Gradient accumulation (rounds >= 2): the reducer must ship the FULLY
+# accumulated per-rank gradient, not round 1's partial one (round-1 bug
+# flagged in VERDICT r01).
+# ---------------------------------------------------------------------------
+
+def _worker_accum(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    from gansformer_amd.parallel.ddp import GradReducer, broadcast_params
+    from gansformer_amd.parallel.dist import setup_distributed
+
+    setup_distributed(backend="gloo")
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.LeakyReLU(),
+        torch.nn.Linear(32, 1))
+    broadcast_params(net)
+    red = GradReducer(net, bucket_mb=0.0001)
+    assert len(red.buckets) >= 2
+
+    rounds = 3
+    red.prepare(rounds)
+    for rnd in range(rounds):
+        torch.manual_seed(1000 + rank * 10 + rnd)
+        x = torch.randn(8, 16)
+        (net(x).sum() / rounds).backward()
+    red.finalize()
+    q.put((rank, net[0].weight.grad.numpy().copy(),
+           net[2].weight.grad.numpy().copy()))
+    dist.destroy_process_group()
+
+
+def _expected_accum(world, rounds=3):
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.LeakyReLU(),
+        torch.nn.Linear(32, 1))
+    sd = {k: v.clone() for k, v in net.state_dict().items()}
+    g0, gl = [], []
+    for rank in range(world):
+        net.load_state_dict(sd)
+        for p in net.parameters():
+            p.grad = None
+        for rnd in range(rounds):
+            torch.manual_seed(1000 + rank * 10 + rnd)
+            x = torch.randn(8, 16)
+            (net(x).sum() / rounds).backward()
+        g0.append(net[0].weight.grad.clone())
+        gl.append(net[2].weight.grad.clone())
+    return torch.stack(g0).mean(0), torch.stack(gl).mean(0)
+
+
+@pytest.mark.timeout(120)
+def test_grad_reducer_accumulation_rounds():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_accum, args=(r, world, 29766, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, g0, gl = q.get(timeout=110)
+        results[rank] = (torch.tensor(g0), torch.tensor(gl))
+    for p in procs:
+        p.join(timeout=30)
+    exp0, expl = _expected_accum(world)
+    for rank in range(world):
+        g0, gl = results[rank]
+        assert torch.allclose(g0, exp0, atol=1e-6), f"rank {rank} accum grads"
+        assert torch.allclose(gl, expl, atol=1e-6), f"rank {rank} accum grads"
+
+
+# ---------------------------------------------------------------------------
+# World-8 dress rehearsal of the FULL trainer step (tiny model): after N
+# steps with per-rank data and gradient accumulation + lazy reg, all ranks
+# must hold bitwise-identical parameters, and sync_ranks() must make
+# w_avg / pl_mean / Gs identical too.
+# ---------------------------------------------------------------------------
+
+def _worker_trainer(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    import copy
+    import torch.distributed as dist
+    from gansformer_amd.models.networks import Discriminator, Generator
+    from gansformer_amd.parallel.ddp import broadcast_params
+    from gansformer_amd.parallel.dist import setup_distributed
+    from gansformer_amd.training.trainer import GANTrainer
+
+    setup_distributed(backend="gloo")
+    torch.manual_seed(0)
+    G = Generator(z_dim=32, w_dim=32, img_resolution=16, num_components=4,
+                  transformer="simplex", channel_base=512, channel_max=64,
+                  bf16_res_count=0, mapping_layers=2)
+    D = Discriminator(img_resolution=16, channel_base=512, channel_max=64,
+                      mbstd_group_size=2, bf16_res_count=0)
+    Gs = copy.deepcopy(G).eval()
+    for p in Gs.parameters():
+        p.requires_grad_(False)
+    broadcast_params(G)
+    broadcast_params(D)
+    broadcast_params(Gs)
+
+    dev = torch.device("cpu")
+    batch_gpu, rounds = 2, 2
+    batch_size = batch_gpu * world * rounds
+    tr = GANTrainer(G, D, Gs, dev, batch_gpu, batch_size,
+                    d_reg_interval=2, g_reg_interval=3, rounds=rounds,
+                    bucket_mb=0.25)
+    torch.manual_seed(5000 + rank)  # per-rank data + z draws
+
+    def next_real():
+        return torch.randn(batch_gpu, 3, 16, 16)
+
+    nimg = 0
+    for step in range(4):  # hits d_reg at 0,2 and g_reg at 0,3
+        tr.step(next_real, step, nimg)
+        nimg += batch_size
+    tr.sync_ranks()
+
+    vec = torch.cat([p.detach().reshape(-1) for p in G.parameters()]
+                    + [p.detach().reshape(-1) for p in D.parameters()])
+    svec = torch.cat([p.detach().reshape(-1) for p in Gs.parameters()]
+                     + [b.detach().reshape(-1) for b in Gs.buffers()])
+    aux = torch.cat([G.mapping.w_avg.reshape(-1),
+                     tr.pl_reg.pl_mean.reshape(-1)])
+    q.put((rank, vec.numpy().copy(), svec.numpy().copy(),
+           aux.numpy().copy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_trainer_world8_rank_identical():
+    world = 8
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_trainer, args=(r, world, 29767, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, vec, svec, aux = q.get(timeout=400)
+        results[rank] = (torch.tensor(vec), torch.tensor(svec),
+                         torch.tensor(aux))
+    for p in procs:
+        p.join(timeout=30)
+    v0, s0, a0 = results[0]
+    assert torch.isfinite(v0).all()
+    for rank in range(1, world):
+        v, s, a = results[rank]
+        assert torch.equal(v, v0), f"rank {rank} params diverged"
+        assert torch.equal(s, s0), f"rank {rank} Gs diverged after sync"
+        assert torch.equal(a, a0), f"rank {rank} w_avg/pl_mean diverged"
+
+
+# ---------------------------------------------------------------------------
+# Multi-rank FID feature gather (gloo): the gathered feature matrix must
+# contain every rank's features (the old path crashed
+# under RCCL; this exercises the same code over gloo).
+# ---------------------------------------------------------------------------
+
+def _worker_fid(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    from gansformer_amd.metrics.fid import collect_features
+    from gansformer_amd.parallel.dist import setup_distributed
+
+    setup_distributed(backend="gloo")
+    dev = torch.device("cpu")
+
+    class ConstFeat(torch.nn.Module):
+        def forward(self, x):
+            return torch.full((x.shape[0], 4), float(rank) + x.mean() * 0)
+
+    def batch_fn(n):
+        return torch.zeros(n, 3, 8, 8)
+
+    feats = collect_features(batch_fn, ConstFeat(), num_images=12,
+                             batch_size=4, device=dev, rank=rank,
+                             world_size=world)
+    q.put((rank, feats.copy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_fid_feature_gather_two_ranks():
+    import numpy as np
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_fid, args=(r, world, 29768, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, feats = q.get(timeout=110)
+        results[rank] = feats
+    for p in procs:
+        p.join(timeout=30)
+    for rank in range(world):
+        f = results[rank]
+        assert f.shape == (12, 4)
+        vals = set(np.unique(f).tolist())
+        assert vals == {0.0, 1.0}, f"gather missing a rank's features: {vals}"
