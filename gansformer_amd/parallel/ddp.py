@@ -42,10 +42,14 @@ class _Bucket:
 
 class GradReducer:
     def __init__(self, module: torch.nn.Module, bucket_mb: float = 25.0,
-                 process_group=None):
+                 process_group=None, force_enabled: bool = False):
+        """force_enabled drives the bucket/collective path even at
+        world_size 1 — used by the single-GPU RCCL smoke test (gpurun
+        boxes have one GPU and RCCL forbids 2 ranks per device)."""
         self.module = module
         self.group = process_group
-        self.enabled = dist.is_initialized() and dist.get_world_size() > 1
+        self.enabled = (dist.is_initialized()
+                        and (dist.get_world_size() > 1 or force_enabled))
         self.world_size = dist.get_world_size() if self.enabled else 1
         self.buckets: list[_Bucket] = []
         self.param_bucket = {}
