@@ -73,12 +73,14 @@ def _out_size(w, u, d, p0, p1, fw):
 
 import weakref
 
-# Keyed on a weakref to the filter TENSOR OBJECT (the resample filters
-# are long-lived registered buffers, so hits are by identity with no
-# host copy). A data_ptr key would alias a freed tensor's reused
-# address and silently return the wrong factorization; a value key
-# would cost a device sync per call. Entries die with their tensor.
-_sep_cache: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
+# Keyed on id() of the filter TENSOR OBJECT with a weakref eviction
+# callback (the resample filters are long-lived registered buffers, so
+# hits are by identity with no host copy). A data_ptr key would alias a
+# freed tensor's reused address and silently return the wrong
+# factorization; a value key would cost a device sync per call.
+# WeakKeyDictionary is unusable here: its lookups fall back to
+# tensor.__eq__, which raises on multi-element tensors.
+_sep_cache: dict = {}  # id(f) -> (weakref(f), result)
 
 
 def _separable8(f):
@@ -86,10 +88,9 @@ def _separable8(f):
     else None."""
     if f.shape != (4, 4):
         return None
-    try:
-        return _sep_cache[f]
-    except KeyError:
-        pass
+    ent = _sep_cache.get(id(f))
+    if ent is not None and ent[0]() is f:
+        return ent[1]
     res = None
     fc = f.detach().cpu()
     r = int(fc.abs().sum(1).argmax())
@@ -100,10 +101,12 @@ def _separable8(f):
         fx = fc[r, :]
         if torch.allclose(torch.outer(fy, fx), fc, atol=1e-7, rtol=1e-5):
             res = torch.cat([fy, fx]).to(f.device)
-    try:
-        _sep_cache[f] = res
-    except TypeError:  # tensor subclass that refuses weakrefs
-        pass
+    key = id(f)
+
+    def _evict(_wr, _key=key):
+        _sep_cache.pop(_key, None)
+
+    _sep_cache[key] = (weakref.ref(f, _evict), res)
     return res
 
 
