@@ -260,6 +260,7 @@ __global__ void attn_longn_reduce(T* __restrict__ out,
                                   const float* __restrict__ ws_m,
                                   const float* __restrict__ ws_l,
                                   const float* __restrict__ ws_o,
+                                  float* __restrict__ ml,  // optional [B,Nq,2]
                                   AttnParams p, int nchunks) {
   const int qq = blockIdx.x;
   const int b = blockIdx.y;
@@ -292,6 +293,11 @@ __global__ void attn_longn_reduce(T* __restrict__ out,
   }
   if (t == 0) l_glob = red[0];
   __syncthreads();
+  if (t == 0 && ml != nullptr) {
+    // saved for the fused backward: A = exp(S*scale - m) / l
+    ml[((long)b * p.Nq + qq) * 2 + 0] = m_glob;
+    ml[((long)b * p.Nq + qq) * 2 + 1] = l_glob;
+  }
   float inv = 1.f / l_glob;
   for (int e = t; e < p.E; e += 256) {
     float acc = 0.f;
@@ -313,14 +319,14 @@ void launch_attn_smalln(T* out, const T* q, const T* k, const T* v,
 }
 template <typename T>
 void launch_attn_longn(T* out, const T* q, const T* k, const T* v,
-                       float* ws_m, float* ws_l, float* ws_o,
+                       float* ws_m, float* ws_l, float* ws_o, float* ml,
                        const AttnParams& p, int nchunks, hipStream_t s) {
   dim3 grid1(nchunks, p.B);
   hipLaunchKernelGGL(attn_longn_partial<T>, grid1, dim3(256), 0, s, ws_m,
                      ws_l, ws_o, q, k, v, p);
   dim3 grid2(p.Nq, p.B);
   hipLaunchKernelGGL(attn_longn_reduce<T>, grid2, dim3(256), 0, s, out, ws_m,
-                     ws_l, ws_o, p, nchunks);
+                     ws_l, ws_o, ml, p, nchunks);
 }
 
 template void launch_attn_smalln<float>(float*, const float*, const float*,
@@ -331,9 +337,11 @@ template void launch_attn_smalln<bf16>(bf16*, const bf16*, const bf16*,
                                        hipStream_t);
 template void launch_attn_longn<float>(float*, const float*, const float*,
                                        const float*, float*, float*, float*,
-                                       const AttnParams&, int, hipStream_t);
+                                       float*, const AttnParams&, int,
+                                       hipStream_t);
 template void launch_attn_longn<bf16>(bf16*, const bf16*, const bf16*,
                                       const bf16*, float*, float*, float*,
-                                      const AttnParams&, int, hipStream_t);
+                                      float*, const AttnParams&, int,
+                                      hipStream_t);
 
 }  // namespace gfa

@@ -84,7 +84,19 @@ void launch_attn_smalln(T*, const T*, const T*, const T*, const AttnParams&,
                         hipStream_t);
 template <typename T>
 void launch_attn_longn(T*, const T*, const T*, const T*, float*, float*,
-                       float*, const AttnParams&, int, hipStream_t);
+                       float*, float*, const AttnParams&, int, hipStream_t);
+struct AttnBwdParams {
+  int B, Nq, Nk, D, E;
+  float scale;
+};
+template <typename T>
+void launch_attn_smalln_bwd(T*, float*, float*, const T*, const T*, const T*,
+                            const T*, const float*, const AttnBwdParams&,
+                            hipStream_t);
+template <typename T>
+void launch_attn_longn_bwd(float*, T*, T*, const T*, const T*, const T*,
+                           const T*, const float*, const float*,
+                           const AttnBwdParams&, hipStream_t);
 
 }  // namespace gfa
 
@@ -417,7 +429,8 @@ Tensor mbstd(Tensor x, int64_t G, int64_t F, double eps) {
   return out;
 }
 
-Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
+std::vector<Tensor> bipartite_attn_impl(Tensor q, Tensor k, Tensor v,
+                                        double scale, bool want_ml) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 3);
   TORCH_CHECK(k.is_cuda() && k.is_contiguous() && k.dim() == 3);
   TORCH_CHECK(v.is_cuda() && v.is_contiguous() && v.dim() == 3);
@@ -434,29 +447,99 @@ Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
   TORCH_CHECK(k.size(0) == p.B && v.size(0) == p.B);
   TORCH_CHECK(k.size(2) == p.D && v.size(1) == p.Nk);
   auto out = torch::empty({p.B, p.Nq, p.E}, q.options());
+  auto f32opt = q.options().dtype(torch::kFloat32);
+  Tensor ml;
   if (p.Nk <= 64) {
     DISPATCH_FT(q, "attn", {
       gfa::launch_attn_smalln<scalar_t>(ptr<scalar_t>(out), cptr<scalar_t>(q),
                                         cptr<scalar_t>(k), cptr<scalar_t>(v),
                                         p, cur_stream());
     });
+    // small-N backward recomputes softmax in-kernel: no stats needed
+    if (want_ml) ml = torch::empty({0}, f32opt);
   } else {
     TORCH_CHECK(p.Nq <= 64,
                 "bipartite_attn: long-N path requires Nq <= 64 "
                 "(bipartite attention is k x HW, never HW x HW)");
     int nchunks = (p.Nk + 63) / 64;
-    auto f32opt = q.options().dtype(torch::kFloat32);
     auto ws_m = torch::empty({p.B, nchunks, 64}, f32opt);
     auto ws_l = torch::empty({p.B, nchunks, 64}, f32opt);
     auto ws_o = torch::empty({(int64_t)p.B, nchunks, 64, p.E}, f32opt);
+    if (want_ml) ml = torch::empty({p.B, p.Nq, 2}, f32opt);
     DISPATCH_FT(q, "attn", {
       gfa::launch_attn_longn<scalar_t>(
           ptr<scalar_t>(out), cptr<scalar_t>(q), cptr<scalar_t>(k),
           cptr<scalar_t>(v), ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
-          ws_o.data_ptr<float>(), p, nchunks, cur_stream());
+          ws_o.data_ptr<float>(),
+          want_ml ? ml.data_ptr<float>() : nullptr, p, nchunks,
+          cur_stream());
     });
   }
-  return out;
+  if (want_ml) return {out, ml};
+  return {out};
+}
+
+Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
+  return bipartite_attn_impl(q, k, v, scale, false)[0];
+}
+
+std::vector<Tensor> bipartite_attn_fwd(Tensor q, Tensor k, Tensor v,
+                                       double scale) {
+  return bipartite_attn_impl(q, k, v, scale, true);
+}
+
+std::vector<Tensor> bipartite_attn_bwd(Tensor q, Tensor k, Tensor v,
+                                       Tensor dout, Tensor drow, Tensor ml,
+                                       double scale) {
+  // drow = (dO * O).sum(-1) fp32 [B,Nq]; ml = [B,Nq,2] fp32 (long-N only).
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 3);
+  TORCH_CHECK(k.is_cuda() && k.is_contiguous());
+  TORCH_CHECK(v.is_cuda() && v.is_contiguous());
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous() && dout.dim() == 3);
+  TORCH_CHECK(drow.is_cuda() && drow.is_contiguous() &&
+              drow.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(q.scalar_type() == k.scalar_type() &&
+              q.scalar_type() == v.scalar_type() &&
+              q.scalar_type() == dout.scalar_type());
+  check_dtype(q, "bipartite_attn_bwd.q");
+  gfa::AttnBwdParams p;
+  p.B = (int)q.size(0);
+  p.Nq = (int)q.size(1);
+  p.Nk = (int)k.size(1);
+  p.D = (int)q.size(2);
+  p.E = (int)v.size(2);
+  p.scale = (float)scale;
+  TORCH_CHECK(dout.size(1) == p.Nq && dout.size(2) == p.E);
+  TORCH_CHECK(drow.size(0) == p.B && drow.size(1) == p.Nq);
+  auto f32opt = q.options().dtype(torch::kFloat32);
+  if (p.Nk <= 64) {
+    auto dq = torch::empty({p.B, p.Nq, p.D}, q.options());
+    auto dkw = torch::zeros({p.B, p.Nk, p.D}, f32opt);
+    auto dvw = torch::zeros({p.B, p.Nk, p.E}, f32opt);
+    DISPATCH_FT(q, "attn_bwd", {
+      gfa::launch_attn_smalln_bwd<scalar_t>(
+          ptr<scalar_t>(dq), dkw.data_ptr<float>(), dvw.data_ptr<float>(),
+          cptr<scalar_t>(q), cptr<scalar_t>(k), cptr<scalar_t>(v),
+          cptr<scalar_t>(dout), drow.data_ptr<float>(), p, cur_stream());
+    });
+    return {dq, dkw.to(q.scalar_type()), dvw.to(q.scalar_type())};
+  }
+  TORCH_CHECK(p.Nq <= 64, "bipartite_attn_bwd: long-N needs Nq <= 64");
+  TORCH_CHECK(ml.numel() == (int64_t)p.B * p.Nq * 2 && ml.is_cuda() &&
+                  ml.is_contiguous() &&
+                  ml.scalar_type() == torch::kFloat32,
+              "bipartite_attn_bwd: long-N needs the forward's (m,l) stats");
+  auto dqw = torch::zeros({p.B, p.Nq, p.D}, f32opt);
+  auto dk = torch::empty({p.B, p.Nk, p.D}, q.options());
+  auto dv = torch::empty({p.B, p.Nk, p.E}, q.options());
+  DISPATCH_FT(q, "attn_bwd", {
+    gfa::launch_attn_longn_bwd<scalar_t>(
+        dqw.data_ptr<float>(), ptr<scalar_t>(dk), ptr<scalar_t>(dv),
+        cptr<scalar_t>(q), cptr<scalar_t>(k), cptr<scalar_t>(v),
+        cptr<scalar_t>(dout), ml.data_ptr<float>(), drow.data_ptr<float>(),
+        p, cur_stream());
+  });
+  return {dqw.to(q.scalar_type()), dk, dv};
 }
 
 }  // namespace
@@ -479,4 +562,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("modnorm_bwd", &modnorm_bwd,
         "first-order modnorm backward (dx, dgamma)");
   m.def("bipartite_attn", &bipartite_attn, "softmax(QK^T)V");
+  m.def("bipartite_attn_fwd", &bipartite_attn_fwd,
+        "softmax(QK^T)V returning (out, ml) for the fused backward");
+  m.def("bipartite_attn_bwd", &bipartite_attn_bwd,
+        "fused attention backward -> (dq, dk, dv)");
 }

@@ -40,18 +40,34 @@ def _eager_attention(q, k, v, scale):
 class _BipartiteAttn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
-        ctx.save_for_backward(q, k, v)
         ctx.scale = scale
-        if native.use_native(q, k, v):
-            return native.require_ext().bipartite_attn(
-                q.contiguous(), k.contiguous(), v.contiguous(), scale)
-        return _eager_attention(q, k, v, scale)[0]
+        ctx.native = native.use_native(q, k, v)
+        if ctx.native:
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+            out, ml = native.require_ext().bipartite_attn_fwd(q, k, v, scale)
+            ctx.save_for_backward(q, k, v, out, ml)
+            return out
+        out, _ = _eager_attention(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, out,
+                              torch.empty(0, device=q.device))
+        return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, = ctx.saved_tensors
+        q, k, v, out, ml = ctx.saved_tensors
         scale = ctx.scale
-        # Recompute A with differentiable ops (double-backward-capable).
+        if ctx.native and not torch.is_grad_enabled():
+            # Fused backward kernels (attn_bwd.hip): recompute A in-LDS
+            # (small-N) or rebuild it from the saved softmax stats
+            # (long-N); drow uses the flash identity
+            # rowsum(dA*A) == rowsum(dO*O).
+            dout_c = dout.contiguous()
+            drow = (dout_c.float() * out.float()).sum(-1).contiguous()
+            dq, dk, dv = native.require_ext().bipartite_attn_bwd(
+                q, k, v, dout_c, drow, ml, scale)
+            return dq, dk, dv, None
+        # create_graph replay (path-length reg) or eager/CPU path:
+        # recompute A with differentiable ops (double-backward-capable).
         # Logits/softmax stay fp32 (the numerics policy the forward
         # kernel implements); the surrounding GEMMs run in the input
         # dtype — hipBLASLt accumulates fp32 internally, and this avoids

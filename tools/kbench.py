@@ -93,6 +93,50 @@ def bench_attn(dtype=torch.bfloat16):
               f"{fl / t / 1e12:7.1f} TF")
 
 
+def bench_attn_bwd(dtype=torch.bfloat16):
+    """Fused attention backward vs the eager recompute composition."""
+    dev = "cuda:0"
+    import math as _m
+    from gansformer_amd.ops import bipartite as bp
+    shapes = [
+        (32, 4096, 17, 512, 512, "simplex res64 b32"),
+        (32, 16384, 17, 256, 256, "simplex res128 b32"),
+        (32, 17, 16384, 256, 256, "duplex-rev res128 b32"),
+        (32, 17, 4096, 512, 512, "duplex-rev res64 b32"),
+    ]
+    for B, Nq, Nk, D, E, note in shapes:
+        scale = 1.0 / _m.sqrt(D)
+        q = torch.randn(B, Nq, D, device=dev, dtype=dtype)
+        k = torch.randn(B, Nk, D, device=dev, dtype=dtype)
+        v = torch.randn(B, Nk, E, device=dev, dtype=dtype)
+        dout = torch.randn(B, Nq, E, device=dev, dtype=dtype)
+        out, ml = C.bipartite_attn_fwd(q, k, v, scale)
+
+        def fused():
+            drow = (dout.float() * out.float()).sum(-1).contiguous()
+            return C.bipartite_attn_bwd(q, k, v, dout, drow, ml, scale)
+
+        def eager():
+            qf, kf = q.float(), k.float()
+            s = torch.einsum("bqd,bkd->bqk", qf, kf) * scale
+            a = torch.softmax(s, dim=-1)
+            a_lp = a.to(v.dtype)
+            dv = torch.einsum("bqk,bqe->bke", a_lp, dout)
+            daf = torch.einsum("bqe,bke->bqk", dout, v).float()
+            ds = a * (daf - (daf * a).sum(dim=-1, keepdim=True))
+            ds_lp = ds.to(q.dtype)
+            dq = torch.einsum("bqk,bkd->bqd", ds_lp, k) * scale
+            dk = torch.einsum("bqk,bqd->bkd", ds_lp, q) * scale
+            return dq, dk, dv
+
+        tf = timeit(fused)
+        te = timeit(eager)
+        fl = 5.0 * B * Nq * Nk * (D + E)  # 5 GEMMs vs fwd's 2
+        print(f"attnbwd  {note:22s}: fused {tf * 1e3:8.3f} ms "
+              f"({fl / tf / 1e12:6.1f} TF)  eager {te * 1e3:8.3f} ms "
+              f"({te / tf:4.1f}x)")
+
+
 def bench_upfirdn(dtype=torch.bfloat16):
     dev = "cuda:0"
     from gansformer_amd.ops.upfirdn2d import setup_filter
@@ -146,5 +190,7 @@ if __name__ == "__main__":
         bench_wgrad()
     if which in ("attn", "all"):
         bench_attn()
+    if which in ("attnbwd", "all"):
+        bench_attn_bwd()
     if which in ("upfirdn", "all"):
         bench_upfirdn()
