@@ -97,6 +97,13 @@ template <typename T>
 void launch_attn_longn_bwd(float*, T*, T*, const T*, const T*, const T*,
                            const T*, const float*, const float*,
                            const AttnBwdParams&, hipStream_t);
+struct GemmParams {
+  long M;
+  int N, K;
+};
+template <typename T>
+void launch_gemm_skinny(T*, const T*, const T*, const GemmParams&, bool,
+                        hipStream_t);
 
 }  // namespace gfa
 
@@ -483,6 +490,31 @@ Tensor bipartite_attn(Tensor q, Tensor k, Tensor v, double scale) {
   return bipartite_attn_impl(q, k, v, scale, false)[0];
 }
 
+Tensor gemm_skinny(Tensor a, Tensor b, bool trans_b) {
+  // C[M,N] = A[M,K] . op(B); trans_b: B is [N,K] (y = x W^T), else [K,N].
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && a.dim() == 2);
+  TORCH_CHECK(b.is_cuda() && b.is_contiguous() && b.dim() == 2);
+  TORCH_CHECK(a.scalar_type() == b.scalar_type());
+  check_dtype(a, "gemm_skinny.a");
+  gfa::GemmParams p;
+  p.M = a.size(0);
+  p.K = (int)a.size(1);
+  if (trans_b) {
+    TORCH_CHECK(b.size(1) == p.K, "gemm_skinny: B [N,K] K mismatch");
+    p.N = (int)b.size(0);
+  } else {
+    TORCH_CHECK(b.size(0) == p.K, "gemm_skinny: B [K,N] K mismatch");
+    p.N = (int)b.size(1);
+  }
+  auto c = torch::empty({p.M, p.N}, a.options());
+  DISPATCH_FT(a, "gemm_skinny", {
+    gfa::launch_gemm_skinny<scalar_t>(ptr<scalar_t>(c), cptr<scalar_t>(a),
+                                      cptr<scalar_t>(b), p, trans_b,
+                                      cur_stream());
+  });
+  return c;
+}
+
 std::vector<Tensor> bipartite_attn_fwd(Tensor q, Tensor k, Tensor v,
                                        double scale) {
   return bipartite_attn_impl(q, k, v, scale, true);
@@ -566,4 +598,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "softmax(QK^T)V returning (out, ml) for the fused backward");
   m.def("bipartite_attn_bwd", &bipartite_attn_bwd,
         "fused attention backward -> (dq, dk, dv)");
+  m.def("gemm_skinny", &gemm_skinny,
+        "tall-skinny C = A @ op(B) MFMA GEMM");
 }

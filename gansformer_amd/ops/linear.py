@@ -13,8 +13,22 @@ from __future__ import annotations
 
 import torch
 
+from . import native
+
 _SPLITK_MIN_M = 65536   # below this the direct GEMM is fine
 _CHUNK = 16384          # target K-chunk per split
+_SKINNY_MIN_M = 65536   # custom MFMA GEMM pays off for tall-skinny shapes
+_SKINNY_MAX_N = 512
+_SKINNY_MAX_K = 1024
+
+
+def _use_skinny(x2, n, k):
+    """Tall-skinny regime where gemm_skinny beats hipBLASLt's ~70 TF/s
+    algorithm picks (profiles/r01_summary.md)."""
+    return (x2.is_cuda and x2.shape[0] >= _SKINNY_MIN_M
+            and n <= _SKINNY_MAX_N and k <= _SKINNY_MAX_K
+            and x2.dtype in (torch.bfloat16, torch.float32)
+            and native.use_native(x2))
 
 
 def _splitk_tn(a, b):
@@ -31,11 +45,18 @@ def _splitk_tn(a, b):
 
 
 class _LinearSplitK(torch.autograd.Function):
-    """y = x @ w^T for x [..., K], w [N, K]."""
+    """y = x @ w^T for x [..., K], w [N, K]. Forward and dx run the
+    tall-skinny MFMA kernel when the shape qualifies; dw stays on the
+    split-K torch composition (already 200-770 TF/s)."""
 
     @staticmethod
     def forward(ctx, x, w):
         ctx.save_for_backward(x, w)
+        x2 = x.reshape(-1, x.shape[-1])
+        if _use_skinny(x2, w.shape[0], w.shape[1]) and w.is_contiguous():
+            y = native.require_ext().gemm_skinny(x2.contiguous(),
+                                                 w, True)
+            return y.reshape(*x.shape[:-1], w.shape[0])
         return x.matmul(w.t())
 
     @staticmethod
@@ -43,7 +64,15 @@ class _LinearSplitK(torch.autograd.Function):
         x, w = ctx.saved_tensors
         dx = dw = None
         if ctx.needs_input_grad[0]:
-            dx = dy.matmul(w)
+            dy2 = dy.reshape(-1, dy.shape[-1])
+            if (not torch.is_grad_enabled()
+                    and _use_skinny(dy2, w.shape[1], w.shape[0])
+                    and w.is_contiguous()):
+                # dx = dy @ W : B operand is [K=N_out, N=K_in] row-major
+                dx = native.require_ext().gemm_skinny(
+                    dy2.contiguous(), w, False).reshape(x.shape)
+            else:
+                dx = dy.matmul(w)
         if ctx.needs_input_grad[1]:
             dyf = dy.reshape(-1, dy.shape[-1])
             xf = x.reshape(-1, x.shape[-1])

@@ -137,6 +137,32 @@ def bench_attn_bwd(dtype=torch.bfloat16):
               f"({te / tf:4.1f}x)")
 
 
+def bench_gemm_skinny(dtype=torch.bfloat16):
+    """Custom tall-skinny MFMA GEMM vs hipBLASLt (x @ w^T)."""
+    dev = "cuda:0"
+    shapes = [
+        (32 * 4096, 512, 512, "proj res64 d512"),
+        (32 * 16384, 256, 256, "proj res128 d256"),
+        (32 * 65536, 128, 128, "proj res256 d128"),
+        (32 * 4096, 128, 512, "gamma res64"),
+        (32 * 65536, 512, 128, "gamma res256"),
+    ]
+    for M, N, K, note in shapes:
+        x = torch.randn(M, K, device=dev, dtype=dtype)
+        w = torch.randn(N, K, device=dev, dtype=dtype)
+        tk = timeit(lambda: C.gemm_skinny(x, w, True))
+        tl = timeit(lambda: x.matmul(w.t()))
+        fl = 2.0 * M * N * K
+        print(f"gemmsk   {note:22s}: ours {tk * 1e3:8.3f} ms "
+              f"({fl / tk / 1e12:6.1f} TF)  blaslt {tl * 1e3:8.3f} ms "
+              f"({fl / tl / 1e12:6.1f} TF)")
+        # dgrad layout (B = [K,N] row-major)
+        wt = w.t().contiguous()
+        tk2 = timeit(lambda: C.gemm_skinny(x, wt, False))
+        print(f"gemmskT  {note:22s}: ours {tk2 * 1e3:8.3f} ms "
+              f"({fl / tk2 / 1e12:6.1f} TF)")
+
+
 def bench_upfirdn(dtype=torch.bfloat16):
     dev = "cuda:0"
     from gansformer_amd.ops.upfirdn2d import setup_filter
@@ -192,5 +218,7 @@ if __name__ == "__main__":
         bench_attn()
     if which in ("attnbwd", "all"):
         bench_attn_bwd()
+    if which in ("gemmsk", "all"):
+        bench_gemm_skinny()
     if which in ("upfirdn", "all"):
         bench_upfirdn()
