@@ -509,3 +509,26 @@ def test_linear_skinny_route_matches(dev):
     close(y, yr, torch.float32)
     close(dx, g @ w.detach(), torch.float32)
     close(dw, g.t() @ x.detach(), torch.float32)
+
+
+def test_mbstd_training_path_uses_kernel(dev):
+    """minibatch_stddev under grad: kernel forward + analytic backward
+    must match the eager composition's grads, including an R1-style
+    double-backward replay."""
+    from gansformer_amd.ops.mbstd import minibatch_stddev, _eager_stats
+    torch.manual_seed(13)
+    x = torch.randn(8, 16, 4, 4, device=dev, requires_grad=True)
+    y = minibatch_stddev(x, group_size=4, num_channels=2)
+    g = torch.randn_like(y)
+    (dx,) = torch.autograd.grad(y, x, g, create_graph=True)
+    # reference: eager stats composition
+    xr = x.detach().clone().requires_grad_(True)
+    stats = _eager_stats(xr, 4, 2, 1e-8)
+    maps = stats.reshape(1, -1, 2, 1, 1).expand(4, 2, 2, 4, 4)
+    yr = torch.cat([xr, maps.reshape(8, 2, 4, 4).to(xr.dtype)], dim=1)
+    (dxr,) = torch.autograd.grad(yr, xr, g, create_graph=True)
+    close(dx, dxr, torch.float32)
+    # double backward (R1 flows through D incl. this layer)
+    dx.square().sum().backward()
+    dxr.square().sum().backward()
+    close(x.grad, xr.grad, torch.float32)

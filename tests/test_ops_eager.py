@@ -296,3 +296,22 @@ def test_linear_splitk_matches_matmul():
     assert gradcheck(_LinearTransposedOut.apply, (u, w), eps=1e-6, atol=1e-4)
     assert gradgradcheck(_LinearTransposedOut.apply, (u, w), eps=1e-6,
                          atol=1e-4)
+
+
+def test_mbstd_analytic_backward_matches_autograd():
+    """_MbStdStats.backward's closed form == autograd through the eager
+    stats composition (CPU, fp64-ish check in fp32)."""
+    import types
+    from gansformer_amd.ops.mbstd import _MbStdStats, _eager_stats
+    torch.manual_seed(3)
+    B, C, H, W, G, F = 8, 6, 5, 5, 4, 2
+    x = torch.randn(B, C, H, W, dtype=torch.float32, requires_grad=True)
+    eps = 1e-8
+    stats = _eager_stats(x, G, F, eps)
+    dstats = torch.randn_like(stats)
+    (dx_ref,) = torch.autograd.grad(stats, x, dstats)
+
+    ctx = types.SimpleNamespace(saved_tensors=(x.detach(),),
+                                params=(G, F, eps))
+    dx = _MbStdStats.backward(ctx, dstats)[0]
+    assert torch.allclose(dx, dx_ref, atol=1e-6, rtol=1e-5)
