@@ -11,14 +11,18 @@
 //
 //   C[o][p]  +=  sum_{t,c} W[o][t][c] * slab[(py+ty(t))(px+tx(t))][c]
 //
-// Weights are passed pre-permuted as [O][9][I] so A-fragments are
-// contiguous 16-B reads. A-fragments are read straight from GLOBAL
-// memory (they are shared by every pixel tile and every sample, so they
-// live in L2: ~8 KB/tap/WG against a 34 TB/s aggregate L2). That leaves
-// the slab as the only LDS producer, so the tap loop runs with NO
-// barriers; one __syncthreads per 32-channel block. The next channel
-// block's slab is prefetched into the other buffer during the MFMA of
-// the current one.
+// Weights are passed pre-packed in the BLOCKED layout
+//   [mTile][cbTile][tap][128 o][32 c]   (o zero-padded to 128)
+// so one wave's A fragment set for a (tap, cb) is a CONTIGUOUS 1 KB
+// region: lanes 0..15 read rows o..o+15 at 16 B each, the four K
+// granules the adjacent columns. The r01 [O][9][I] layout made each
+// fragment a 16-way 16-B gather with stride 9*I — the strided L2
+// traffic that bounded the kernel (A-constant experiment: 1169 vs 725
+// TF/s, profiles/r01_summary.md). A-fragments still come from global
+// (L2-hot), so the slab stays the only LDS producer and the tap loop
+// runs with NO barriers; one __syncthreads per 32-channel block. The
+// next channel block's slab is prefetched into the other buffer during
+// the MFMA of the current one.
 //
 // Tiles (template TH): TH=16 -> 128(O) x 256(px as 16x16), each wave a
 // 32x256 sub-tile (2x16 fragments; every A fragment reused 16x, no O row
@@ -43,7 +47,7 @@ constexpr int SLAB_PIX = 40;           // u16 per slab pixel (32 + 8 pad)
 template <int TH>
 __global__ __launch_bounds__(256, TH == 16 ? 1 : 2)
 void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
-                          const bf16* __restrict__ wr,  // [O][9][I]
+                          const bf16* __restrict__ wr,  // blocked, see top
                           int B, int I, int H, int W, int O) {
   // TH=8: wave = 64(O) x 64(px), acc 4x4. TH=16: wave = 32(O) x 256(px),
   // acc 2x16 — each A fragment reused 16x and no row is loaded by two
@@ -71,19 +75,20 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
   const int px = lane & 15;               // B-fragment pixel x (fixed)
   const int ag = lane >> 4;               // K granule 0..3
 
-  // A-fragment global bases for this lane's 4 M rows (+tap*I+c0 later).
-  // Out-of-range O rows are CLAMPED, not masked: they compute garbage
-  // partial sums whose stores the epilogue skips, so no zero-fill or
-  // per-fragment predicate is needed (saves registers). Offsets fit
-  // int32 (O*9*I <= 2.4M elements).
+  const int nCB = I >> 5;
+
+  // A-fragment bases in the blocked layout: row inside this m-tile's
+  // 128-padded block (out-of-range rows are ZERO-padded by the packer,
+  // so no clamp/predicate needed) plus the K granule column. The
+  // (cb, tap) block offset is added in the loop: each tap block is
+  // 128*32 = 4096 elements. Offsets fit int32 (mT*nCB*9*4096 <= 2.4M
+  // + padding).
   int a_off[MI];
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
-    const int o = min(m0 + wm + mi * 16 + (lane & 15), O - 1);
-    a_off[mi] = o * 9 * I + ag * 8;
+    const int orow = wm + mi * 16 + (lane & 15);
+    a_off[mi] = blockIdx.y * nCB * 9 * 4096 + orow * 32 + ag * 8;
   }
-
-  const int nCB = I >> 5;
 
   // Slab staging is split into a LOAD phase (global -> registers, issued
   // at the top of a channel block) and a WRITE phase (registers -> LDS,
@@ -189,14 +194,14 @@ void conv2d_fwd_slab_bf16(bf16* __restrict__ y, const bf16* __restrict__ x,
 #pragma unroll
     for (int tap = 0; tap < 9; ++tap) {
       const int r = tap / 3, s = tap - r * 3;
-      const int tc = tap * I + c0;
+      const int tc = (cb * 9 + tap) * 4096;
       s16x8 af[MI];
       if (TH == 16) {
 #pragma unroll
         for (int mi = 0; mi < MI; ++mi) af[mi] = afp[mi];
         const bool more = tap < 8 || pre;
         if (more) {
-          const int ntc = tap < 8 ? tc + I : c0 + 32;
+          const int ntc = tap < 8 ? tc + 4096 : (cb + 1) * 9 * 4096;
 #pragma unroll
           for (int mi = 0; mi < MI; ++mi)
             afp[mi] = *reinterpret_cast<const s16x8*>(wr + a_off[mi] + ntc);

@@ -289,10 +289,19 @@ Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
                                cptr<float>(w), p, cur_stream());
   } else if (gfa::conv2d_slab_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
                                        p.kw, p.stride, p.pad, p.per_sample)) {
-    // tap-major slab kernel wants weights as [O][kh*kw][I]
-    auto wr = w.permute({0, 2, 3, 1}).reshape({O, kh * kw, I}).contiguous();
+    // blocked weight layout [mT][cB][9][128][32] (o zero-padded to 128):
+    // one wave's A fragment set per (tap, cb) is a contiguous 1 KB read
+    int mT = (O + 127) / 128, cB = I / 32;
+    Tensor wsrc = w;
+    if (O % 128 != 0)
+      wsrc = torch::constant_pad_nd(w.reshape({O, (int64_t)I * 9}),
+                                    {0, 0, 0, (int64_t)mT * 128 - O})
+                 .reshape({(int64_t)mT * 128, I, 3, 3});
+    auto wb = wsrc.reshape({mT, 128, cB, 32, 9})
+                  .permute({0, 2, 4, 1, 3})
+                  .contiguous();
     gfa::launch_conv2d_fwd_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
-                                     cptr<gfa::bf16>(wr), p.B, p.I, p.H, p.W,
+                                     cptr<gfa::bf16>(wb), p.B, p.I, p.H, p.W,
                                      p.O, cur_stream());
   } else if (gfa::conv2d_s2_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
                                      p.kw, p.stride, p.pad, p.per_sample)) {
