@@ -19,8 +19,11 @@
 //    (m, l) softmax stats, writes its own dK/dV rows directly, and
 //    atomically accumulates dQ (small: [Nq, D]).
 //
-// fp32 softmax/dS math in both paths; MFMA operands in the input dtype
-// (bf16 path: v_mfma_f32_16x16x32_bf16), accumulate fp32.
+// All global staging is vectorized: bf16 rows move as 16-B s16x8
+// granules (raw bit copies, no per-element float round-trips) — the
+// first version's scalar 2-B loads made the kernel pure-issue-bound
+// (~680 load instructions per thread). fp32 softmax/dS math in both
+// paths; MFMA operands in the input dtype, accumulate fp32.
 #include "common.h"
 
 namespace gfa {
@@ -46,6 +49,7 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   constexpr int BK = TO::BK;
   constexpr int ROW = TO::ROW;
   using elem = typename TO::elem;
+  constexpr bool BF = (sizeof(elem) == 2);
   __shared__ float Ss[64][68];          // A, later overwritten with dS
   __shared__ elem Ta[64 * 2 * ROW];     // staging pair (depth up to 64)
   __shared__ elem Tb[64 * 2 * ROW];
@@ -56,6 +60,10 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   const int lane = t & 63;
   const int wave = t >> 6;
   const int nk_t16 = (p.Nk + 15) / 16;
+  const int nk_bk = (p.Nk <= BK) ? 1 : 2;  // depth tiles when depth = keys
+  // D/E are attention dims (multiples of 8 in every model config); the
+  // vector path needs 16-B aligned rows
+  const bool vec8 = BF && (p.D % 8 == 0) && (p.E % 8 == 0);
 
   const T* qb = q + ((long)b * p.Nq) * p.D;
   const T* kb = k + ((long)b * p.Nk) * p.D;
@@ -66,19 +74,64 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   float* dvb = dvw + ((long)b * p.Nk) * p.E;
   const float* drb = drow + (long)b * p.Nq;
 
+  // stage a [64 rows x BK] operand tile: rows from row-major global
+  // (stride rstride), columns [c0, c0+BK) clipped at cmax, rows >= vrows
+  // zero-filled.
+  auto stage_rows = [&](elem* buf, const T* src, long rstride, int vrows,
+                        int c0, int cmax) {
+    if constexpr (BF) {
+      if (vec8) {
+        const int row = t >> 2, g = t & 3;  // 256 tasks exactly
+        const int c = c0 + g * 8;
+        s16x8 val{};
+        if (row < vrows && c < cmax)
+          val = *reinterpret_cast<const s16x8*>(src + row * rstride + c);
+        TO::store_vec8(buf, row, g, val);
+        return;
+      }
+    }
+    for (int i = t; i < 64 * BK; i += 256) {
+      int row = i / BK, c = c0 + i % BK;
+      float v_ = 0.f;
+      if (row < vrows && c < cmax) v_ = to_f32(src[row * rstride + c]);
+      TO::store(buf, row, i % BK, v_);
+    }
+  };
+
+  // stage a transposed [64 global-cols x 64 src-rows] tile pair:
+  // buf rows = global column index (c0 + 0..63), depth = source row.
+  auto stage_trans = [&](elem* buf, const T* src, long rstride, int vrows,
+                         int c0, int cmax) {
+    if constexpr (BF) {
+      if (vec8) {
+        for (int task = t; task < 512; task += 256) {
+          const int srow = task >> 3, g = task & 7;
+          const int c = c0 + g * 8;
+          s16x8 val{};
+          if (srow < vrows && c < cmax)
+            val = *reinterpret_cast<const s16x8*>(src + srow * rstride + c);
+          const u16* pv = reinterpret_cast<const u16*>(&val);
+          elem* tb = buf + (srow / BK) * 64 * ROW;
+          const int kk = srow % BK;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) tb[(g * 8 + j) * ROW + kk] = pv[j];
+        }
+        return;
+      }
+    }
+    for (int i = t; i < 64 * 64; i += 256) {
+      int srow = i >> 6, c = c0 + (i & 63);
+      float v_ = 0.f;
+      if (srow < vrows && c < cmax) v_ = to_f32(src[srow * rstride + c]);
+      TO::store(buf + (srow / BK) * 64 * ROW, i & 63, srow % BK, v_);
+    }
+  };
+
   // ---------- phase A: S = Q K^T -> softmax -> A in Ss ----------
   f32x4 acc_s[4] = {};
   for (int d0 = 0; d0 < p.D; d0 += BK) {
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, dd = d0 + i % BK;
-      float qv = 0.f, kv = 0.f;
-      if (dd < p.D) {
-        if (q0 + row < p.Nq) qv = to_f32(qb[(long)(q0 + row) * p.D + dd]);
-        if (row < p.Nk) kv = to_f32(kb[(long)row * p.D + dd]);
-      }
-      TO::store(Ta, row, i % BK, qv);
-      TO::store(Tb, row, i % BK, kv);
-    }
+    stage_rows(Ta, qb + (long)q0 * p.D, p.D, p.Nq - q0, d0, p.D);
+    stage_rows(Tb, kb, p.D, p.Nk, d0, p.D);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -114,14 +167,7 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
     TO::store(Ta + (qq / BK) * 64 * ROW, key, qq % BK, Ss[qq][key]);
   }
   for (int e0 = 0; e0 < p.E; e0 += 64) {
-    // dO^T tile: rows = e, depth = q (coalesced over e)
-    for (int i = t; i < 64 * 64; i += 256) {
-      int qq = i >> 6, e = i & 63;
-      float dv_ = 0.f;
-      if (q0 + qq < p.Nq && e0 + e < p.E)
-        dv_ = to_f32(dob[(long)(q0 + qq) * p.E + e0 + e]);
-      TO::store(Tb + (qq / BK) * 64 * ROW, e, qq % BK, dv_);
-    }
+    stage_trans(Tb, dob + (long)q0 * p.E, p.E, p.Nq - q0, e0, p.E);
     __syncthreads();
     f32x4 accv[4] = {};
 #pragma unroll
@@ -146,16 +192,8 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   // ---------- phase C: dA = dO V^T; dS = A.*(dA - drow)*scale ----------
   f32x4 acc_da[4] = {};
   for (int e0 = 0; e0 < p.E; e0 += BK) {
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, ee = e0 + i % BK;
-      float dov = 0.f, vv = 0.f;
-      if (ee < p.E) {
-        if (q0 + row < p.Nq) dov = to_f32(dob[(long)(q0 + row) * p.E + ee]);
-        if (row < p.Nk) vv = to_f32(vb[(long)row * p.E + ee]);
-      }
-      TO::store(Ta, row, i % BK, dov);
-      TO::store(Tb, row, i % BK, vv);
-    }
+    stage_rows(Ta, dob + (long)q0 * p.E, p.E, p.Nq - q0, e0, p.E);
+    stage_rows(Tb, vb, p.E, p.Nk, e0, p.E);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -181,10 +219,7 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
     for (int reg = 0; reg < 4; ++reg) {
       int key = ni * 16 + (lane & 15);
       int qq = wave * 16 + (lane >> 4) * 4 + reg;
-      if (ni < nk_t16)
-        Ss[qq][key] = acc_da[ni][reg];
-      else if (key < 64)
-        Ss[qq][key] = 0.f;
+      Ss[qq][key] = (ni < nk_t16) ? acc_da[ni][reg] : 0.f;
     }
   __syncthreads();
 
@@ -194,19 +229,12 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
     TO::store(Ta + (key / BK) * 64 * ROW, qq, key % BK, Ss[qq][key]);
   }
   for (int d0 = 0; d0 < p.D; d0 += 64) {
-    // K^T tile: rows = d, depth = key (coalesced over d)
-    for (int i = t; i < 64 * 64; i += 256) {
-      int key = i >> 6, dd = i & 63;
-      float kv = 0.f;
-      if (key < p.Nk && d0 + dd < p.D)
-        kv = to_f32(kb[(long)key * p.D + d0 + dd]);
-      TO::store(Tb + (key / BK) * 64 * ROW, dd, key % BK, kv);
-    }
+    stage_trans(Tb, kb, p.D, p.Nk, d0, p.D);  // K^T: rows = d, depth = key
     __syncthreads();
     f32x4 accq[4] = {};
 #pragma unroll
     for (int di = 0; di < 4; ++di)
-      for (int kk = 0; kk < 64; kk += BK)
+      for (int kk = 0; kk < nk_bk * BK; kk += BK)
         accq[di] = TO::mfma(Ta + (kk / BK) * 64 * ROW,
                             Tb + (kk / BK) * 64 * ROW, wave * 16, di * 16,
                             lane, accq[di]);
@@ -229,14 +257,7 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
     TO::store(Ta + (qq / BK) * 64 * ROW, key, qq % BK, Ss[qq][key]);
   }
   for (int d0 = 0; d0 < p.D; d0 += 64) {
-    // Q^T tile: rows = d, depth = q
-    for (int i = t; i < 64 * 64; i += 256) {
-      int qq = i >> 6, dd = i & 63;
-      float qv = 0.f;
-      if (q0 + qq < p.Nq && d0 + dd < p.D)
-        qv = to_f32(qb[(long)(q0 + qq) * p.D + d0 + dd]);
-      TO::store(Tb + (qq / BK) * 64 * ROW, dd, qq % BK, qv);
-    }
+    stage_trans(Tb, qb + (long)q0 * p.D, p.D, p.Nq - q0, d0, p.D);
     __syncthreads();
     f32x4 acck[4] = {};
 #pragma unroll
@@ -276,6 +297,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
   constexpr int BK = TO::BK;
   constexpr int ROW = TO::ROW;
   using elem = typename TO::elem;
+  constexpr bool BF = (sizeof(elem) == 2);
   __shared__ float Ss[64][68];          // A_c, later dS_c ([key][q])
   __shared__ elem Ta[64 * 2 * ROW];
   __shared__ elem Tb[64 * 2 * ROW];
@@ -286,6 +308,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
   const int lane = t & 63;
   const int wave = t >> 6;
   const int nq_t16 = (p.Nq + 15) / 16;
+  const bool vec8 = BF && (p.D % 8 == 0) && (p.E % 8 == 0);
 
   const T* qb = q + ((long)b * p.Nq) * p.D;
   const T* kb = k + ((long)b * p.Nk) * p.D;
@@ -295,19 +318,58 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
   T* dkb = dk + ((long)b * p.Nk) * p.D;
   T* dvb = dv + ((long)b * p.Nk) * p.E;
 
+  auto stage_rows = [&](elem* buf, const T* src, long rstride, int vrows,
+                        int c0, int cmax) {
+    if constexpr (BF) {
+      if (vec8) {
+        const int row = t >> 2, g = t & 3;
+        const int c = c0 + g * 8;
+        s16x8 val{};
+        if (row < vrows && c < cmax)
+          val = *reinterpret_cast<const s16x8*>(src + row * rstride + c);
+        TO::store_vec8(buf, row, g, val);
+        return;
+      }
+    }
+    for (int i = t; i < 64 * BK; i += 256) {
+      int row = i / BK, c = c0 + i % BK;
+      float v_ = 0.f;
+      if (row < vrows && c < cmax) v_ = to_f32(src[row * rstride + c]);
+      TO::store(buf, row, i % BK, v_);
+    }
+  };
+  auto stage_trans = [&](elem* buf, const T* src, long rstride, int vrows,
+                         int c0, int cmax) {
+    if constexpr (BF) {
+      if (vec8) {
+        for (int task = t; task < 512; task += 256) {
+          const int srow = task >> 3, g = task & 7;
+          const int c = c0 + g * 8;
+          s16x8 val{};
+          if (srow < vrows && c < cmax)
+            val = *reinterpret_cast<const s16x8*>(src + srow * rstride + c);
+          const u16* pv = reinterpret_cast<const u16*>(&val);
+          elem* tb = buf + (srow / BK) * 64 * ROW;
+          const int kk = srow % BK;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) tb[(g * 8 + j) * ROW + kk] = pv[j];
+        }
+        return;
+      }
+    }
+    for (int i = t; i < 64 * 64; i += 256) {
+      int srow = i >> 6, c = c0 + (i & 63);
+      float v_ = 0.f;
+      if (srow < vrows && c < cmax) v_ = to_f32(src[srow * rstride + c]);
+      TO::store(buf + (srow / BK) * 64 * ROW, i & 63, srow % BK, v_);
+    }
+  };
+
   // ---------- phase A: S_c = K_c Q^T -> A_c via saved (m, l) ----------
   f32x4 acc_s[4] = {};
   for (int d0 = 0; d0 < p.D; d0 += BK) {
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, dd = d0 + i % BK;
-      float kv = 0.f, qv = 0.f;
-      if (dd < p.D) {
-        if (key0 + row < p.Nk) kv = to_f32(kb[(long)(key0 + row) * p.D + dd]);
-        if (row < p.Nq) qv = to_f32(qb[(long)row * p.D + dd]);
-      }
-      TO::store(Ta, row, i % BK, kv);
-      TO::store(Tb, row, i % BK, qv);
-    }
+    stage_rows(Ta, kb + (long)key0 * p.D, p.D, p.Nk - key0, d0, p.D);
+    stage_rows(Tb, qb, p.D, p.Nq, d0, p.D);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -321,8 +383,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
     for (int reg = 0; reg < 4; ++reg) {
       int qq = ni * 16 + (lane & 15);
       int key = wave * 16 + (lane >> 4) * 4 + reg;
-      if (qq < 64)
-        Ss[key][qq] = (ni < nq_t16) ? acc_s[ni][reg] : 0.f;
+      Ss[key][qq] = (ni < nq_t16) ? acc_s[ni][reg] : 0.f;
     }
   __syncthreads();
   for (int i = t; i < 64 * 64; i += 256) {
@@ -343,13 +404,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
     TO::store(Ta + (qq / BK) * 64 * ROW, key, qq % BK, Ss[key][qq]);
   }
   for (int e0 = 0; e0 < p.E; e0 += 64) {
-    for (int i = t; i < 64 * 64; i += 256) {
-      int qq = i >> 6, e = i & 63;
-      float dv_ = 0.f;
-      if (qq < p.Nq && e0 + e < p.E)
-        dv_ = to_f32(dob[(long)qq * p.E + e0 + e]);
-      TO::store(Tb + (qq / BK) * 64 * ROW, e, qq % BK, dv_);
-    }
+    stage_trans(Tb, dob, p.E, p.Nq, e0, p.E);
     __syncthreads();
     f32x4 accv[4] = {};
 #pragma unroll
@@ -374,16 +429,8 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
   // ---------- phase C: dA_c = V_c dO^T; dS_c = A.*(dA-drow)*scale ----------
   f32x4 acc_da[4] = {};
   for (int e0 = 0; e0 < p.E; e0 += BK) {
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, ee = e0 + i % BK;
-      float vv = 0.f, dov = 0.f;
-      if (ee < p.E) {
-        if (key0 + row < p.Nk) vv = to_f32(vb[(long)(key0 + row) * p.E + ee]);
-        if (row < p.Nq) dov = to_f32(dob[(long)row * p.E + ee]);
-      }
-      TO::store(Ta, row, i % BK, vv);
-      TO::store(Tb, row, i % BK, dov);
-    }
+    stage_rows(Ta, vb + (long)key0 * p.E, p.E, p.Nk - key0, e0, p.E);
+    stage_rows(Tb, dob, p.E, p.Nq, e0, p.E);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -409,8 +456,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
     for (int reg = 0; reg < 4; ++reg) {
       int qq = ni * 16 + (lane & 15);
       int key = wave * 16 + (lane >> 4) * 4 + reg;
-      if (qq < 64)
-        Ss[key][qq] = (ni < nq_t16) ? acc_da[ni][reg] : 0.f;
+      Ss[key][qq] = (ni < nq_t16) ? acc_da[ni][reg] : 0.f;
     }
   __syncthreads();
 
@@ -420,13 +466,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
     TO::store(Ta + (qq / BK) * 64 * ROW, key, qq % BK, Ss[key][qq]);
   }
   for (int d0 = 0; d0 < p.D; d0 += 64) {
-    for (int i = t; i < 64 * 64; i += 256) {
-      int qq = i >> 6, dd = i & 63;
-      float qv = 0.f;
-      if (qq < p.Nq && d0 + dd < p.D)
-        qv = to_f32(qb[(long)qq * p.D + d0 + dd]);
-      TO::store(Tb + (qq / BK) * 64 * ROW, dd, qq % BK, qv);
-    }
+    stage_trans(Tb, qb, p.D, p.Nq, d0, p.D);  // Q^T: rows = d, depth = q
     __syncthreads();
     f32x4 acck[4] = {};
 #pragma unroll
@@ -454,14 +494,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
     TO::store(Ta + (key / BK) * 64 * ROW, qq, key % BK, Ss[key][qq]);
   }
   for (int d0 = 0; d0 < p.D; d0 += 64) {
-    // K_c^T tile: rows = d, depth = key
-    for (int i = t; i < 64 * 64; i += 256) {
-      int key = i >> 6, dd = i & 63;
-      float kv = 0.f;
-      if (key0 + key < p.Nk && d0 + dd < p.D)
-        kv = to_f32(kb[(long)(key0 + key) * p.D + d0 + dd]);
-      TO::store(Tb + (key / BK) * 64 * ROW, dd, key % BK, kv);
-    }
+    stage_trans(Tb, kb + (long)key0 * p.D, p.D, p.Nk - key0, d0, p.D);
     __syncthreads();
     f32x4 accq[4] = {};
 #pragma unroll
