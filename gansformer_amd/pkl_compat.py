@@ -143,6 +143,24 @@ def build_from_state(state: dict) -> torch.nn.Module:
 
 def load_variables(module: torch.nn.Module, variables, strict=True):
     sd = module.state_dict()
+    variables = list(variables)
+    # Reference pickles carry TF graph-scoped names (G_synthesis/8x8/...)
+    # with TF shape conventions; route those through the best-effort name
+    # map (tf_name_map.py) instead of failing on every variable.
+    n_scoped = sum(1 for n, _ in variables if "/" in n)
+    if variables and n_scoped > len(variables) // 2:
+        from .tf_name_map import map_tf_variables
+        top_res = getattr(module, "img_resolution", None)
+        mapped, unmapped = map_tf_variables(variables, sd, top_res=top_res)
+        for name, arr in mapped.items():
+            t = torch.as_tensor(np.array(arr))
+            sd[name].copy_(t.to(sd[name].dtype).reshape(sd[name].shape))
+        if unmapped and strict:
+            raise KeyError(
+                f"{len(unmapped)} TF-scoped pkl variables could not be "
+                f"mapped: {unmapped[:5]}... (pass strict=False to load "
+                f"the mapped subset)")
+        return unmapped
     missing = []
     for name, arr in variables:
         if name in sd:

@@ -78,3 +78,78 @@ def test_unpickler_maps_tflib_network_to_stub():
     obj = pkl_compat._RestrictedUnpickler(io.BytesIO(payload)).load()
     assert isinstance(obj, pkl_compat.NetworkStub)
     assert obj.state["name"] == "G"
+
+
+def test_tf_scoped_variable_loading():
+    """A TF-style variables list (reference naming + TF shape
+    conventions [R]) loads through the name map with correct
+    transposition."""
+    import numpy as np
+    from gansformer_amd.models.networks import Discriminator, Generator
+    from gansformer_amd.pkl_compat import load_variables
+
+    torch.manual_seed(0)
+    G = Generator(z_dim=32, w_dim=32, img_resolution=16, num_components=4,
+                  transformer="none", channel_base=512, channel_max=64,
+                  bf16_res_count=0, mapping_layers=2)
+    sd = {k: v.clone() for k, v in G.state_dict().items()}
+    tf_vars = []
+    # mapping denses (TF [in,out])
+    for j in range(2):
+        w = sd[f"mapping.layers.{j}.weight"].numpy()
+        tf_vars.append((f"G_mapping/Dense{j}/weight", w.T.copy() + 1.0))
+        tf_vars.append((f"G_mapping/Dense{j}/bias",
+                        sd[f"mapping.layers.{j}.bias"].numpy() + 1.0))
+    # one synthesis conv (TF [kh,kw,in,out])
+    w = sd["synthesis.blocks.1.conv0.weight"].numpy()
+    tf_vars.append(("G_synthesis/8x8/Conv0_up/weight",
+                    np.transpose(w, (2, 3, 1, 0)) + 1.0))
+    tf_vars.append(("G_synthesis/8x8/Conv0_up/mod_weight",
+                    sd["synthesis.blocks.1.conv0.affine.weight"].numpy().T
+                    + 1.0))
+    tf_vars.append(("G_synthesis/4x4/Const/const",
+                    sd["synthesis.blocks.0.const"].numpy() + 1.0))
+
+    unmapped = load_variables(G, tf_vars, strict=True)
+    assert unmapped == []
+    sd2 = G.state_dict()
+    for our in ("mapping.layers.0.weight", "mapping.layers.1.bias",
+                "synthesis.blocks.1.conv0.weight",
+                "synthesis.blocks.1.conv0.affine.weight",
+                "synthesis.blocks.0.const"):
+        assert torch.allclose(sd2[our], sd[our] + 1.0), our
+
+    # D-side round trip
+    D = Discriminator(img_resolution=16, channel_base=512, channel_max=64,
+                      mbstd_group_size=2, bf16_res_count=0)
+    dsd = {k: v.clone() for k, v in D.state_dict().items()}
+    wd = dsd["blocks.0.conv1.weight"].numpy()
+    tf_d = [
+        ("D/16x16/FromRGB/weight",
+         np.transpose(dsd["frgb.weight"].numpy(), (2, 3, 1, 0)) + 1.0),
+        ("D/16x16/Conv1_down/weight", np.transpose(wd, (2, 3, 1, 0)) + 1.0),
+        ("D/16x16/Skip/weight",
+         np.transpose(dsd["blocks.0.skip.weight"].numpy(),
+                      (2, 3, 1, 0)) + 1.0),
+        ("D/4x4/Dense0/weight", dsd["fc.weight"].numpy().T + 1.0),
+        ("D/Output/bias", dsd["out.bias"].numpy() + 1.0),
+    ]
+    assert load_variables(D, tf_d, strict=True) == []
+    dsd2 = D.state_dict()
+    assert torch.allclose(dsd2["blocks.0.conv1.weight"],
+                          dsd["blocks.0.conv1.weight"] + 1.0)
+    assert torch.allclose(dsd2["fc.weight"], dsd["fc.weight"] + 1.0)
+
+
+def test_tf_scoped_unmapped_raises():
+    from gansformer_amd.models.networks import Generator
+    from gansformer_amd.pkl_compat import load_variables
+    import numpy as np
+    G = Generator(z_dim=32, w_dim=32, img_resolution=16, num_components=4,
+                  transformer="none", channel_base=512, channel_max=64,
+                  bf16_res_count=0, mapping_layers=2)
+    bogus = [(f"G_synthesis/8x8/Mystery/weight", np.zeros((2, 2)))
+             for _ in range(3)]
+    with pytest.raises(KeyError):
+        load_variables(G, bogus, strict=True)
+    assert len(load_variables(G, bogus, strict=False)) == 3
