@@ -122,3 +122,24 @@ def test_generator_integration_variants(integration):
     img.square().mean().backward()
     assert all(p.grad is None or torch.isfinite(p.grad).all()
                for p in G.parameters())
+
+
+def test_down_conv_exact_ordering_small_res():
+    """At feature maps <= 32 the D down-conv uses the exact pad-baked
+    blur->VALID-conv ordering: border pixels must match the reference
+    composition exactly (the reordered form deviated up to
+    0.56 relative on borders where they are 75% of the output)."""
+    from gansformer_amd.models.networks import Conv2dLayer as ConvLayer
+    from gansformer_amd.ops import conv2d_gradfix, setup_filter, upfirdn2d
+    torch.manual_seed(0)
+    for H in (8, 16, 32):
+        layer = ConvLayer(8, 16, kernel_size=3, act="linear", down=2,
+                          bias=False)
+        x = torch.randn(2, 8, H, H)
+        y = layer(x)
+        f = setup_filter([1, 3, 3, 1])
+        w = (layer.weight * layer.weight_gain)
+        xe = upfirdn2d(x, f, padding=(2, 3, 2, 3))
+        ye = conv2d_gradfix(xe, w, stride=2, padding=0)
+        assert y.shape == ye.shape
+        assert torch.allclose(y, ye, atol=1e-5), f"H={H}"
