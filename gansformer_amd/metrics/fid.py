@@ -1,19 +1,27 @@
-"""FID harness (ref src/metrics/frechet_inception_distance.py [R]).
+"""FID harness (ref src/metrics/frechet_inference_distance.py [R]).
 
 Frechet distance between feature distributions of reals and generated
-images, with cached real statistics (mu, Sigma) like the reference. The
-canonical extractor (torchvision Inception-v3 with pretrained weights) is
-NOT available in this offline environment, so the harness is
-extractor-pluggable:
+images, with cached real statistics (mu, Sigma) like the reference.
+
+EXTRACTOR PROTOCOL. The canonical extractor (Inception-v3, pretrained)
+cannot exist in this offline environment: torchvision/timm are not
+installed, no weight cache exists on the image, and there is no
+network egress (verified 2026-09-13; see BASELINE.md "FID protocol").
+The harness is therefore extractor-pluggable:
 
   * `--inception-path` / `extractor=` : a TorchScript module or
-    state-dict path producing [B, F] features — use this to get true
-    FID-50k numbers when weights are present.
-  * default: `RandomConvFeatures` — a frozen, seed-deterministic conv
-    net (random features are a known, usable FID family; numbers are
-    comparable between runs of THIS framework, not to published
-    Inception FIDs).
+    state-dict path producing [B, F] features — drop in an Inception-v3
+    artifact to get literature-comparable FID-50k numbers.
+  * default: **FID-RC/v1** (`RandomConvFeatures`, feature_dim=2048,
+    seed=123, architecture pinned below): a frozen, seed-deterministic
+    random conv net. Random-feature FID is a valid distance family
+    (it responds monotonically to distribution drift); FID-RC numbers
+    are comparable ONLY between runs of this framework using the same
+    protocol version, never to published Inception FIDs. Any change to
+    the architecture/seed below MUST bump the protocol name.
 
+Metric files record the protocol (`metric-fid*.txt` lines carry the
+extractor tag) so cross-round numbers are always same-extractor A/B.
 The Frechet-distance math (the judged "FID harness correctness") is
 exact and unit-tested against closed-form Gaussian cases.
 """
@@ -28,8 +36,13 @@ import torch.nn as nn
 import torch.distributed as tdist
 
 
+FID_PROTOCOL = "FID-RC/v1"  # random-conv extractor protocol version
+
+
 class RandomConvFeatures(nn.Module):
-    """Frozen random conv feature extractor, deterministic in `seed`."""
+    """Frozen random conv feature extractor, deterministic in `seed`.
+    Architecture + defaults are pinned as protocol FID-RC/v1 — do not
+    change without bumping FID_PROTOCOL (see module docstring)."""
 
     def __init__(self, feature_dim=2048, seed=123):
         super().__init__()
@@ -54,6 +67,13 @@ class RandomConvFeatures(nn.Module):
     def forward(self, x):
         # x: float images in [-1, 1]
         return self.net(x).flatten(1)
+
+
+def extractor_tag(path_or_none):
+    """Protocol tag recorded next to every FID number."""
+    if path_or_none and os.path.exists(path_or_none):
+        return os.path.basename(path_or_none)
+    return FID_PROTOCOL
 
 
 def load_extractor(path_or_none, device):

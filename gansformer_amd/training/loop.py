@@ -177,8 +177,11 @@ def training_loop(
             real_stats_cache[name] = stats
             results[name] = fid
             if is_main and run_dir is not None:
+                from ..metrics.fid import extractor_tag
+                tag = extractor_tag(dataset_args.get("inception_path"))
                 with open(os.path.join(run_dir, f"metric-{name}.txt"), "a") as f:
-                    f.write(f"{cur_nimg // 1000:d}kimg {fid:.4f}\n")
+                    f.write(f"{cur_nimg // 1000:d}kimg {fid:.4f} "
+                            f"extractor={tag}\n")
         return results
 
     cur_nimg = start_nimg
