@@ -70,7 +70,8 @@ def test_training_loop_metrics(tmp_path):
     assert os.path.exists(tmp_path / "metric-fid1k.txt")
     with open(tmp_path / "metric-fid1k.txt") as f:
         line = f.read().strip()
-    assert "kimg" in line and float(line.split()[-1]) >= 0
+    assert "kimg" in line and float(line.split()[1]) >= 0
+    assert "extractor=" in line  # same-extractor A/B protocol tag
 
 
 def test_training_loop_grad_accumulation(tmp_path):
