@@ -532,3 +532,26 @@ def test_mbstd_training_path_uses_kernel(dev):
     dx.square().sum().backward()
     dxr.square().sum().backward()
     close(x.grad, xr.grad, torch.float32)
+
+
+# ---------------- determinism (LDS race detection) ----------------
+def test_slab_kernels_deterministic(dev):
+    """The slab kernels elide barriers in the tap loop (csrc/README.md
+    'Race-freedom'). LDS races show up as run-to-run nondeterminism
+    under wave-scheduling jitter: run each kernel repeatedly on the
+    same inputs and require bitwise-identical outputs."""
+    torch.manual_seed(14)
+    cases = []
+    x1 = torch.randn(3, 64, 32, 32, device=dev, dtype=torch.bfloat16)
+    w1 = torch.randn(160, 64, 3, 3, device=dev, dtype=torch.bfloat16) * 0.1
+    cases.append(lambda: _C().conv2d_fwd(x1, w1, 1, 1))
+    x2 = torch.randn(2, 64, 64, 64, device=dev, dtype=torch.bfloat16)
+    w2 = torch.randn(128, 64, 3, 3, device=dev, dtype=torch.bfloat16) * 0.1
+    cases.append(lambda: _C().conv2d_fwd(x2, w2, 2, 1))      # s2 slab
+    dy = torch.randn(2, 128, 64, 64, device=dev, dtype=torch.bfloat16)
+    cases.append(lambda: _C().conv2d_wgrad(x2, dy, 1, 1, 3, 3, False))
+    for fn in cases:
+        ref = fn().clone()
+        for _ in range(8):
+            out = fn()
+            assert torch.equal(out, ref), "nondeterministic kernel output"
