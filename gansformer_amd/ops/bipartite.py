@@ -45,22 +45,33 @@ class _BipartiteAttn(torch.autograd.Function):
         if ctx.native:
             q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
             out, ml = native.require_ext().bipartite_attn_fwd(q, k, v, scale)
-            ctx.save_for_backward(q, k, v, out, ml)
+            if k.shape[1] > 64:
+                # long-N: the fused backward needs out (for drow) + the
+                # softmax stats. Small-N uses the eager recompute and
+                # must not pin the [B,HW,E] output through backward.
+                ctx.save_for_backward(q, k, v, out, ml)
+            else:
+                e = torch.empty(0, device=q.device)
+                ctx.save_for_backward(q, k, v, e, e)
             return out
+        e = torch.empty(0, device=q.device)
         out, _ = _eager_attention(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, out,
-                              torch.empty(0, device=q.device))
+        ctx.save_for_backward(q, k, v, e, e)
         return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, ml = ctx.saved_tensors
         scale = ctx.scale
-        if ctx.native and not torch.is_grad_enabled():
-            # Fused backward kernels (attn_bwd.hip): recompute A in-LDS
-            # (small-N) or rebuild it from the saved softmax stats
-            # (long-N); drow uses the flash identity
-            # rowsum(dA*A) == rowsum(dO*O).
+        # Fused backward is routed for the LONG-N direction only
+        # (duplex reverse, softmax over HW): measured 1.5-1.6x over the
+        # eager recompute (gpurun_out/r02_attnbwd2.log). The small-N
+        # fused kernel is correct but 0.7x (its 5-phase LDS dance can't
+        # beat hipBLASLt's 400+ TF/s on the skinny recompute GEMMs), so
+        # small-N keeps the eager composition.
+        if ctx.native and k.shape[1] > 64 and not torch.is_grad_enabled():
+            # rebuild A from the saved softmax stats; drow uses the
+            # flash identity rowsum(dA*A) == rowsum(dO*O).
             dout_c = dout.contiguous()
             drow = (dout_c.float() * out.float()).sum(-1).contiguous()
             dq, dk, dv = native.require_ext().bipartite_attn_bwd(

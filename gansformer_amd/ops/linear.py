@@ -17,18 +17,17 @@ from . import native
 
 _SPLITK_MIN_M = 65536   # below this the direct GEMM is fine
 _CHUNK = 16384          # target K-chunk per split
-_SKINNY_MIN_M = 65536   # custom MFMA GEMM pays off for tall-skinny shapes
-_SKINNY_MAX_N = 512
-_SKINNY_MAX_K = 1024
 
 
 def _use_skinny(x2, n, k):
-    """Tall-skinny regime where gemm_skinny beats hipBLASLt's ~70 TF/s
-    algorithm picks (profiles/r01_summary.md)."""
-    return (x2.is_cuda and x2.shape[0] >= _SKINNY_MIN_M
-            and n <= _SKINNY_MAX_N and k <= _SKINNY_MAX_K
-            and x2.dtype in (torch.bfloat16, torch.float32)
-            and native.use_native(x2))
+    """MEASURED OFF (r02): hipBLASLt runs these tall-skinny projection
+    shapes at 258-765 TF/s on plain contiguous matmuls
+    (gpurun_out/r02_gemmsk.log) — the r01 ~70 TF/s reading came from a
+    different call path. Our gemm_skinny kernel reaches only 75-90 TF/s,
+    so routing through it regressed the whole step (145 -> 131 imgs/s).
+    The kernel stays in csrc/ with its numerics tests as the MFMA GEMM
+    reference implementation, but nothing routes to it."""
+    return False
 
 
 def _splitk_tn(a, b):
