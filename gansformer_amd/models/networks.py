@@ -127,15 +127,17 @@ class Conv2dLayer(nn.Module):
                 # subsample) and keeps tensor sizes even
                 x = downsample2d(x, self.resample_filter, down=self.down)
                 y = conv2d_gradfix(x, w, stride=1, padding=0)
-            elif x.shape[-1] <= 32:
+            elif x.shape[-1] <= 16:
                 # EXACT pad-baked ordering (reference semantics): bake
                 # the conv's padding into the blur so border outputs see
                 # blurred contributions, then a VALID strided conv. At
-                # these sizes border pixels are 23-75% of the output
+                # these sizes border pixels are 44-75% of the output
                 # (measured: ~0.35-0.56 relative border deviation in the
                 # reordered form) and the stride-2 slab kernel is not
-                # eligible anyway, so exactness is free. PARITY.md "D
-                # resample ordering".
+                # eligible anyway (it needs W >= 32), so exactness is
+                # free. At 32 the fast path keeps the s2 slab kernel
+                # (a threshold of 32 cost ~6% whole-step throughput).
+                # PARITY.md "D resample ordering".
                 p0 = (fh - self.down + 1) // 2 + self.padding
                 p1 = fh - 1 - (fh - self.down + 1) // 2 + self.padding
                 x = upfirdn2d(x, self.resample_filter,
@@ -290,26 +292,38 @@ class BipartiteLayer(nn.Module):
 
 class MappingNetwork(nn.Module):
     def __init__(self, z_dim=512, w_dim=512, num_components=16, num_ws=14,
-                 num_layers=8, lr_mul=0.01, w_avg_beta=0.995):
+                 num_layers=8, lr_mul=0.01, w_avg_beta=0.995, label_dim=0):
         super().__init__()
         self.z_dim, self.w_dim = z_dim, w_dim
         self.num_components = num_components  # k (attention latents)
         self.num_latents = num_components + 1  # + global
         self.num_ws = num_ws
         self.w_avg_beta = w_avg_beta
+        self.label_dim = label_dim
+        if label_dim > 0:
+            # class-conditional path (SURVEY M1 "+class embed"): the
+            # label embedding is normalized and concatenated to every
+            # latent component before the shared MLP
+            self.label_embed = FullyConnected(label_dim, z_dim)
         layers = []
-        dims = [z_dim] + [w_dim] * num_layers
+        dims = [z_dim * (2 if label_dim > 0 else 1)] + [w_dim] * num_layers
         for i in range(num_layers):
             layers.append(FullyConnected(dims[i], dims[i + 1], act="lrelu",
                                          lr_mul=lr_mul))
         self.layers = nn.ModuleList(layers)
         self.register_buffer("w_avg", torch.zeros(self.num_latents, w_dim))
 
-    def forward(self, z, truncation_psi=1.0, truncation_cutoff=None,
-                update_w_avg=False):
+    def forward(self, z, label=None, truncation_psi=1.0,
+                truncation_cutoff=None, update_w_avg=False):
         """z: [B, k+1, z_dim] -> ws [B, num_ws, k+1, w_dim]."""
         assert z.ndim == 3 and z.shape[1] == self.num_latents
         x = pixel_norm(z.to(torch.float32), dim=-1)
+        if self.label_dim > 0:
+            assert label is not None and label.shape[-1] == self.label_dim, \
+                "conditional mapping needs a [B, label_dim] label"
+            y = pixel_norm(self.label_embed(label.to(torch.float32)), dim=-1)
+            x = torch.cat([x, y.unsqueeze(1).expand(-1, self.num_latents,
+                                                    -1)], dim=-1)
         for layer in self.layers:
             x = layer(x)
         if update_w_avg:
@@ -495,7 +509,7 @@ class Generator(nn.Module):
                  attn_resolutions=None, use_pos=True, channel_base=32768,
                  channel_max=512, bf16_res_count=16, mapping_layers=8,
                  mapping_lr_mul=0.01, w_avg_beta=0.995, conv_clamp=256.0,
-                 style_mixing_prob=0.9):
+                 style_mixing_prob=0.9, label_dim=0):
         super().__init__()
         self.z_dim = z_dim
         self.w_dim = w_dim
@@ -503,6 +517,7 @@ class Generator(nn.Module):
         self.img_channels = img_channels
         self.num_components = num_components
         self.style_mixing_prob = style_mixing_prob
+        self.label_dim = label_dim
         self.init_kwargs = dict(
             z_dim=z_dim, w_dim=w_dim, img_resolution=img_resolution,
             img_channels=img_channels, num_components=num_components,
@@ -511,7 +526,8 @@ class Generator(nn.Module):
             channel_base=channel_base, channel_max=channel_max,
             bf16_res_count=bf16_res_count, mapping_layers=mapping_layers,
             mapping_lr_mul=mapping_lr_mul, w_avg_beta=w_avg_beta,
-            conv_clamp=conv_clamp, style_mixing_prob=style_mixing_prob)
+            conv_clamp=conv_clamp, style_mixing_prob=style_mixing_prob,
+            label_dim=label_dim)
         self.synthesis = SynthesisNetwork(
             w_dim=w_dim, img_resolution=img_resolution,
             img_channels=img_channels, num_components=num_components,
@@ -522,7 +538,8 @@ class Generator(nn.Module):
         self.mapping = MappingNetwork(
             z_dim=z_dim, w_dim=w_dim, num_components=num_components,
             num_ws=self.synthesis.num_ws, num_layers=mapping_layers,
-            lr_mul=mapping_lr_mul, w_avg_beta=w_avg_beta)
+            lr_mul=mapping_lr_mul, w_avg_beta=w_avg_beta,
+            label_dim=label_dim)
 
     @property
     def num_latents(self):
@@ -532,17 +549,17 @@ class Generator(nn.Module):
         return torch.randn(batch_size, self.num_latents, self.z_dim,
                            device=device, generator=generator)
 
-    def forward(self, z, truncation_psi=1.0, truncation_cutoff=None,
-                noise_mode="random", style_mixing=False, update_w_avg=False,
-                return_ws=False):
-        ws = self.mapping(z, truncation_psi=truncation_psi,
+    def forward(self, z, label=None, truncation_psi=1.0,
+                truncation_cutoff=None, noise_mode="random",
+                style_mixing=False, update_w_avg=False, return_ws=False):
+        ws = self.mapping(z, label=label, truncation_psi=truncation_psi,
                           truncation_cutoff=truncation_cutoff,
                           update_w_avg=update_w_avg)
         if style_mixing and self.style_mixing_prob > 0:
             if float(torch.rand(())) < self.style_mixing_prob:
                 cutoff = int(torch.randint(1, ws.shape[1], ()))
                 z2 = torch.randn_like(z)
-                ws2 = self.mapping(z2)
+                ws2 = self.mapping(z2, label=label)
                 ws = torch.cat([ws[:, :cutoff], ws2[:, cutoff:]], dim=1)
         img = self.synthesis(ws, noise_mode=noise_mode)
         if return_ws:
@@ -576,18 +593,20 @@ class DiscriminatorBlock(nn.Module):
 class Discriminator(nn.Module):
     def __init__(self, img_resolution=256, img_channels=3, channel_base=32768,
                  channel_max=512, mbstd_group_size=4, mbstd_num_channels=1,
-                 bf16_res_count=16, conv_clamp=256.0):
+                 bf16_res_count=16, conv_clamp=256.0, label_dim=0):
         super().__init__()
         self.img_resolution = img_resolution
         self.img_channels = img_channels
         self.mbstd_group_size = mbstd_group_size
         self.mbstd_num_channels = mbstd_num_channels
+        self.label_dim = label_dim
         self.init_kwargs = dict(
             img_resolution=img_resolution, img_channels=img_channels,
             channel_base=channel_base, channel_max=channel_max,
             mbstd_group_size=mbstd_group_size,
             mbstd_num_channels=mbstd_num_channels,
-            bf16_res_count=bf16_res_count, conv_clamp=conv_clamp)
+            bf16_res_count=bf16_res_count, conv_clamp=conv_clamp,
+            label_dim=label_dim)
         res_log2 = int(math.log2(img_resolution))
         self.block_resolutions = [2 ** i for i in range(res_log2, 2, -1)]
         bf16_start = img_resolution / (2 ** (bf16_res_count - 1)) \
@@ -608,8 +627,13 @@ class Discriminator(nn.Module):
                                     act="lrelu", conv_clamp=conv_clamp)
         self.fc = FullyConnected(prev_ch * 4 * 4, prev_ch, act="lrelu")
         self.out = FullyConnected(prev_ch, 1)
+        if label_dim > 0:
+            # projection-discriminator conditioning: the logit gets a
+            # label-embedding dot product with the final features
+            self.label_embed = FullyConnected(label_dim, prev_ch)
+        self._proj_dim = prev_ch
 
-    def forward(self, img, force_fp32=False):
+    def forward(self, img, label=None, force_fp32=False):
         blk0_bf16 = (self.blocks and self.blocks[0].use_bf16
                      and not force_fp32 and img.is_cuda)
         x = self.frgb(img.to(torch.bfloat16 if blk0_bf16 else torch.float32))
@@ -619,4 +643,11 @@ class Discriminator(nn.Module):
         x = minibatch_stddev(x, self.mbstd_group_size, self.mbstd_num_channels)
         x = self.conv_out(x)
         x = self.fc(x.flatten(1))
-        return self.out(x)
+        logit = self.out(x)
+        if self.label_dim > 0:
+            assert label is not None and label.shape[-1] == self.label_dim, \
+                "conditional D needs a [B, label_dim] label"
+            emb = self.label_embed(label.to(x.dtype))
+            logit = logit + (emb * x).sum(dim=-1, keepdim=True) \
+                / math.sqrt(self._proj_dim)
+        return logit

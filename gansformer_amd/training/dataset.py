@@ -183,6 +183,10 @@ class MirroredDataset(Dataset):
     def image_shape(self):
         return self.base.image_shape
 
+    @property
+    def label_dim(self):
+        return getattr(self.base, "label_dim", 0)
+
     def __getitem__(self, idx):
         n = len(self.base)
         img, label = self.base[idx % n]
@@ -192,9 +196,12 @@ class MirroredDataset(Dataset):
 
 
 def load_dataset(dataset=None, data_dir=None, resolution=256, channels=3,
-                 synthetic_size=50000, seed=0, mirror_augment=False):
+                 synthetic_size=50000, seed=0, mirror_augment=False,
+                 label_dim=0):
     """Resolve a dataset spec: a prepared dir if it exists, else
-    synthetic (`dataset="shapes"` selects the structured shapes data)."""
+    synthetic (`dataset="shapes"` selects the structured shapes data).
+    label_dim > 0 makes the synthetic datasets emit one-hot labels
+    (prepared datasets carry their own labels.npy)."""
     ds = None
     if data_dir and dataset:
         path = os.path.join(data_dir, dataset)
@@ -202,10 +209,12 @@ def load_dataset(dataset=None, data_dir=None, resolution=256, channels=3,
             ds = ShardedNpyDataset(path, resolution=resolution)
     if ds is None and dataset in ("shapes", "clevr-synth"):
         ds = ShapesDataset(resolution=resolution, channels=channels,
-                           size=synthetic_size, seed=seed)
+                           size=synthetic_size, seed=seed,
+                           label_dim=label_dim)
     if ds is None:
         ds = SyntheticDataset(resolution=resolution, channels=channels,
-                              size=synthetic_size, seed=seed)
+                              size=synthetic_size, seed=seed,
+                              label_dim=label_dim)
     if mirror_augment:
         ds = MirroredDataset(ds)
     return ds

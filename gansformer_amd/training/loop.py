@@ -83,10 +83,13 @@ def training_loop(
     res = ds.resolution
     chans = ds.image_shape[0]
 
-    # networks
-    G_kwargs = dict(img_resolution=res, img_channels=chans)
+    # networks (conditional iff the dataset carries labels)
+    label_dim = getattr(ds, "label_dim", 0)
+    G_kwargs = dict(img_resolution=res, img_channels=chans,
+                    label_dim=label_dim)
     G_kwargs.update(G_args)
-    D_kwargs = dict(img_resolution=res, img_channels=chans)
+    D_kwargs = dict(img_resolution=res, img_channels=chans,
+                    label_dim=label_dim)
     D_kwargs.update(D_args)
     G = Generator(**G_kwargs).to(dev).train()
     D = Discriminator(**D_kwargs).to(dev).train()
@@ -105,6 +108,17 @@ def training_loop(
     broadcast_params(D)
     broadcast_params(Gs)
 
+    conditional = G_kwargs.get("label_dim", 0) > 0
+
+    def next_real():
+        imgs_u8, labels = next(loader)
+        imgs = normalize_images(imgs_u8, dev)
+        return (imgs, labels.to(dev)) if conditional else imgs
+
+    def sample_labels(n):
+        idx = np.random.randint(0, len(ds), size=n)
+        return torch.stack([ds[int(i)][1] for i in idx]).to(dev)
+
     trainer = GANTrainer(
         G, D, Gs, dev, batch_gpu, batch_size,
         g_lr=sched_args.get("g_lr", 0.002),
@@ -117,7 +131,8 @@ def training_loop(
         g_reg_interval=loss_args.get("g_reg_interval", 8),
         pl_weight=loss_args.get("pl_weight", 2.0),
         pl_decay=loss_args.get("pl_decay", 0.01),
-        ema_kimg=ema_kimg, ema_rampup=ema_rampup, rounds=rounds)
+        ema_kimg=ema_kimg, ema_rampup=ema_rampup, rounds=rounds,
+        label_fn=sample_labels if conditional else None)
     if extra is not None:
         trainer.load_state_dict(extra)
         if "cur_nimg" in extra and not resume_kimg:
@@ -143,10 +158,6 @@ def training_loop(
     metric_extractor = None
     real_stats_cache = {}
 
-    def next_real():
-        imgs_u8, _labels = next(loader)
-        return normalize_images(imgs_u8, dev)
-
     def grab_real_batch(n):
         idx = np.random.randint(0, len(ds), size=n)
         xs = torch.stack([ds[int(i)][0] for i in idx])
@@ -167,7 +178,9 @@ def training_loop(
 
             def gen_batch(n):
                 z = Gs.sample_z(n, device=dev)
-                return Gs(z, truncation_psi=1.0, noise_mode="random")
+                lab = sample_labels(n) if conditional else None
+                return Gs(z, label=lab, truncation_psi=1.0,
+                          noise_mode="random")
 
             fid, stats = compute_fid(
                 gen_batch, grab_real_batch, n_img, min(batch_gpu * 2, 64),
@@ -203,6 +216,7 @@ def training_loop(
         prof.start()
 
     grid_z = G.sample_z(16, device=dev)
+    grid_label = sample_labels(16) if conditional else None
 
     while cur_nimg < total_kimg * 1000:
         trainer.step(next_real, step, cur_nimg, start_nimg=start_nimg,
@@ -232,7 +246,8 @@ def training_loop(
 
         if is_main and run_dir is not None and cur_nimg >= next_img:
             with torch.no_grad():
-                fakes = Gs(grid_z, truncation_psi=0.7, noise_mode="const")
+                fakes = Gs(grid_z, label=grid_label, truncation_psi=0.7,
+                           noise_mode="const")
             snap.save_image_grid(
                 fakes, os.path.join(run_dir, f"fakes{cur_nimg // 1000:06d}.png"))
             next_img = cur_nimg + image_snapshot_kimg * 1000

@@ -19,7 +19,8 @@ class GANTrainer:
                  g_lr=0.002, d_lr=0.002, beta1=0.0, beta2=0.99, eps=1e-8,
                  gamma=10.0, d_reg_interval=16, g_reg_interval=8,
                  pl_weight=2.0, pl_decay=0.01, ema_kimg=10.0,
-                 ema_rampup=0.05, rounds=1, bucket_mb=25.0, r1_fp32=False):
+                 ema_rampup=0.05, rounds=1, bucket_mb=25.0, r1_fp32=False,
+                 label_fn=None):
         self.G, self.D, self.Gs = G, D, Gs
         self.device = device
         self.batch_gpu = batch_gpu
@@ -38,6 +39,12 @@ class GANTrainer:
                                d_reg_interval)
         self.g_red = GradReducer(G, bucket_mb=bucket_mb)
         self.d_red = GradReducer(D, bucket_mb=bucket_mb)
+        # class-conditional training: label_fn(n) -> [n, label_dim]
+        # samples generator labels from the dataset's label marginal
+        self.conditional = getattr(G, "label_dim", 0) > 0
+        self.label_fn = label_fn
+        if self.conditional and label_fn is None:
+            raise ValueError("conditional G (label_dim > 0) needs label_fn")
 
     def step(self, next_real_fn, step_idx, cur_nimg, start_nimg=0,
              logger=None):
@@ -47,17 +54,22 @@ class GANTrainer:
         do_d_reg = self.d_reg_interval > 0 and step_idx % self.d_reg_interval == 0
         do_g_reg = self.g_reg_interval > 0 and step_idx % self.g_reg_interval == 0
         rep = logger.report if logger is not None else (lambda *a: None)
+        cond = self.conditional
+
+        def unpack(r):
+            return r if isinstance(r, (tuple, list)) else (r, None)
 
         # ---------------- D main ----------------
         self.d_opt.zero_grad(set_to_none=True)
         self.d_red.prepare(rounds)
         for _ in range(rounds):
-            real = next_real_fn()
+            real, rlab = unpack(next_real_fn())
             z = G.sample_z(self.batch_gpu, device=self.device)
+            flab = self.label_fn(self.batch_gpu) if cond else None
             with torch.no_grad():
-                fake = G(z, style_mixing=True)
-            real_logits = D(real)
-            fake_logits = D(fake)
+                fake = G(z, label=flab, style_mixing=True)
+            real_logits = D(real, label=rlab) if cond else D(real)
+            fake_logits = D(fake, label=flab) if cond else D(fake)
             loss_d = d_logistic_loss(real_logits, fake_logits) / rounds
             loss_d.backward()
             rep("Loss/D", loss_d.detach() * rounds)
@@ -71,8 +83,10 @@ class GANTrainer:
             self.d_opt.zero_grad(set_to_none=True)
             self.d_red.prepare(rounds)
             for _ in range(rounds):
-                real = next_real_fn().requires_grad_(True)
-                real_logits = D(real, force_fp32=self.r1_fp32)
+                real, rlab = unpack(next_real_fn())
+                real = real.requires_grad_(True)
+                real_logits = D(real, label=rlab, force_fp32=self.r1_fp32) \
+                    if cond else D(real, force_fp32=self.r1_fp32)
                 r1 = r1_penalty(real_logits, real)
                 ((self.gamma * r1 * self.d_reg_interval) / rounds).backward()
                 rep("Loss/r1", r1.detach())
@@ -84,8 +98,9 @@ class GANTrainer:
         self.g_red.prepare(rounds)
         for _ in range(rounds):
             z = G.sample_z(self.batch_gpu, device=self.device)
-            fake = G(z, style_mixing=True, update_w_avg=True)
-            fake_logits = D(fake)
+            flab = self.label_fn(self.batch_gpu) if cond else None
+            fake = G(z, label=flab, style_mixing=True, update_w_avg=True)
+            fake_logits = D(fake, label=flab) if cond else D(fake)
             loss_g = g_nonsaturating_loss(fake_logits) / rounds
             loss_g.backward()
             rep("Loss/G", loss_g.detach() * rounds)
@@ -99,7 +114,8 @@ class GANTrainer:
             pl_batch = max(self.batch_gpu // 2, 1)
             for _ in range(rounds):
                 z = G.sample_z(pl_batch, device=self.device)
-                ws = G.mapping(z)
+                flab = self.label_fn(pl_batch) if cond else None
+                ws = G.mapping(z, label=flab)
                 fake = G.synthesis(ws)
                 pl = self.pl_reg(fake, ws)
                 ((pl * self.g_reg_interval) / rounds).backward()

@@ -84,3 +84,13 @@ def test_training_loop_grad_accumulation(tmp_path):
         rows = [json.loads(l) for l in f if l.strip()]
     assert rows and all(
         torch.isfinite(torch.tensor(r.get("Loss/D", 0.0))) for r in rows)
+
+
+def test_training_loop_conditional(tmp_path):
+    """Dataset with labels -> conditional G/D end to end (class embed in
+    mapping, projection head in D; SURVEY M1 '+class embed')."""
+    out = training_loop(**cfg(
+        tmp_path,
+        dataset_args=dict(dataset="synthetic", resolution=32,
+                          synthetic_size=64, label_dim=5)))
+    assert out["steps"] >= 1

@@ -132,7 +132,7 @@ def test_down_conv_exact_ordering_small_res():
     from gansformer_amd.models.networks import Conv2dLayer as ConvLayer
     from gansformer_amd.ops import conv2d_gradfix, setup_filter, upfirdn2d
     torch.manual_seed(0)
-    for H in (8, 16, 32):
+    for H in (8, 16):
         layer = ConvLayer(8, 16, kernel_size=3, act="linear", down=2,
                           bias=False)
         x = torch.randn(2, 8, H, H)
