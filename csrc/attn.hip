@@ -51,20 +51,14 @@ void attn_smalln_kernel(T* __restrict__ out, const T* __restrict__ q,
   const T* vb = v + ((long)b * p.Nk) * p.E;
   T* ob = out + ((long)b * p.Nq) * p.E;
 
+  const bool vecD = (sizeof(elem) == 2) && (p.D % 8 == 0);
+  const bool vecE = (sizeof(elem) == 2) && (p.E % 8 == 0);
   // ---------- S = Q K^T ----------
   f32x4 acc_s[4] = {};
   for (int d0 = 0; d0 < p.D; d0 += BK) {
-    // stage Q tile rows (64 q) and K tile rows (keys; zero-pad)
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, dd = d0 + i % BK;
-      float qv = 0.f, kv = 0.f;
-      if (dd < p.D) {
-        if (q0 + row < p.Nq) qv = to_f32(qb[(long)(q0 + row) * p.D + dd]);
-        if (row < p.Nk) kv = to_f32(kb[(long)row * p.D + dd]);
-      }
-      TO::store(Qs, row, i % BK, qv);
-      TO::store(Ks, row, i % BK, kv);
-    }
+    stage_tile_rows<T>(t, Qs, qb + (long)q0 * p.D, p.D, p.Nq - q0, d0, p.D,
+                       vecD);
+    stage_tile_rows<T>(t, Ks, kb, p.D, p.Nk, d0, p.D, vecD);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -106,14 +100,8 @@ void attn_smalln_kernel(T* __restrict__ out, const T* __restrict__ q,
   // ---------- O = P V ----------
   const int nk_pad = (p.Nk + BK - 1) / BK * BK;  // BK-multiple key depth
   for (int e0 = 0; e0 < p.E; e0 += 64) {
-    // stage V^T tile: rows = e (64), cols = keys
-    for (int i = t; i < 64 * 64; i += 256) {
-      int key = i >> 6, e = i & 63;  // coalesced over e for fixed key
-      float vv = 0.f;
-      if (key < p.Nk && e0 + e < p.E)
-        vv = to_f32(vb[(long)key * p.E + e0 + e]);
-      TO::store(Vt + (key / BK) * 64 * ROW, e, key % BK, vv);
-    }
+    // stage V^T tile pair: rows = e (64), depth = keys
+    stage_tile_trans<T>(t, Vt, vb, p.E, p.Nk, e0, p.E, vecE);
     __syncthreads();
     f32x4 acc_o[4] = {};
 #pragma unroll
@@ -170,19 +158,14 @@ void attn_longn_partial(float* __restrict__ ws_m, float* __restrict__ ws_l,
   const T* kb = k + ((long)b * p.Nk) * p.D;
   const T* vb = v + ((long)b * p.Nk) * p.E;
 
+  const bool vecD = (sizeof(elem) == 2) && (p.D % 8 == 0);
+  const bool vecE = (sizeof(elem) == 2) && (p.E % 8 == 0);
   // ---------- S_c = K_c Q^T  (rows = keys, cols = queries) ----------
   f32x4 acc_s[4] = {};
   for (int d0 = 0; d0 < p.D; d0 += BK) {
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, dd = d0 + i % BK;
-      float kv = 0.f, qv = 0.f;
-      if (dd < p.D) {
-        if (key0 + row < p.Nk) kv = to_f32(kb[(long)(key0 + row) * p.D + dd]);
-        if (row < p.Nq) qv = to_f32(qb[(long)row * p.D + dd]);
-      }
-      TO::store(Ks, row, i % BK, kv);
-      TO::store(Qs, row, i % BK, qv);
-    }
+    stage_tile_rows<T>(t, Ks, kb + (long)key0 * p.D, p.D, p.Nk - key0, d0,
+                       p.D, vecD);
+    stage_tile_rows<T>(t, Qs, qb, p.D, p.Nq, d0, p.D, vecD);
     __syncthreads();
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
@@ -224,13 +207,8 @@ void attn_longn_partial(float* __restrict__ ws_m, float* __restrict__ ws_l,
   // ---------- O_c = P_c V_c ----------
   float* ob = ws_o + (((long)b * nchunks + c) * 64) * p.E;
   for (int e0 = 0; e0 < p.E; e0 += 64) {
-    for (int i = t; i < 64 * 64; i += 256) {
-      int key = i >> 6, e = i & 63;
-      float vv = 0.f;
-      if (key0 + key < p.Nk && e0 + e < p.E)
-        vv = to_f32(vb[(long)(key0 + key) * p.E + e0 + e]);
-      TO::store(Vt + (key / BK) * 64 * ROW, e, key % BK, vv);
-    }
+    stage_tile_trans<T>(t, Vt, vb + (long)key0 * p.E, p.E, p.Nk - key0, e0,
+                        p.E, vecE);
     __syncthreads();
     f32x4 acc_o[4] = {};
 #pragma unroll

@@ -74,57 +74,14 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   float* dvb = dvw + ((long)b * p.Nk) * p.E;
   const float* drb = drow + (long)b * p.Nq;
 
-  // stage a [64 rows x BK] operand tile: rows from row-major global
-  // (stride rstride), columns [c0, c0+BK) clipped at cmax, rows >= vrows
-  // zero-filled.
+  // staging delegates to the shared helpers in common.h
   auto stage_rows = [&](elem* buf, const T* src, long rstride, int vrows,
                         int c0, int cmax) {
-    if constexpr (BF) {
-      if (vec8) {
-        const int row = t >> 2, g = t & 3;  // 256 tasks exactly
-        const int c = c0 + g * 8;
-        s16x8 val{};
-        if (row < vrows && c < cmax)
-          val = *reinterpret_cast<const s16x8*>(src + row * rstride + c);
-        TO::store_vec8(buf, row, g, val);
-        return;
-      }
-    }
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, c = c0 + i % BK;
-      float v_ = 0.f;
-      if (row < vrows && c < cmax) v_ = to_f32(src[row * rstride + c]);
-      TO::store(buf, row, i % BK, v_);
-    }
+    stage_tile_rows<T>(t, buf, src, rstride, vrows, c0, cmax, vec8);
   };
-
-  // stage a transposed [64 global-cols x 64 src-rows] tile pair:
-  // buf rows = global column index (c0 + 0..63), depth = source row.
   auto stage_trans = [&](elem* buf, const T* src, long rstride, int vrows,
                          int c0, int cmax) {
-    if constexpr (BF) {
-      if (vec8) {
-        for (int task = t; task < 512; task += 256) {
-          const int srow = task >> 3, g = task & 7;
-          const int c = c0 + g * 8;
-          s16x8 val{};
-          if (srow < vrows && c < cmax)
-            val = *reinterpret_cast<const s16x8*>(src + srow * rstride + c);
-          const u16* pv = reinterpret_cast<const u16*>(&val);
-          elem* tb = buf + (srow / BK) * 64 * ROW;
-          const int kk = srow % BK;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) tb[(g * 8 + j) * ROW + kk] = pv[j];
-        }
-        return;
-      }
-    }
-    for (int i = t; i < 64 * 64; i += 256) {
-      int srow = i >> 6, c = c0 + (i & 63);
-      float v_ = 0.f;
-      if (srow < vrows && c < cmax) v_ = to_f32(src[srow * rstride + c]);
-      TO::store(buf + (srow / BK) * 64 * ROW, i & 63, srow % BK, v_);
-    }
+    stage_tile_trans<T>(t, buf, src, rstride, vrows, c0, cmax, vec8);
   };
 
   // ---------- phase A: S = Q K^T -> softmax -> A in Ss ----------
@@ -320,49 +277,11 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
 
   auto stage_rows = [&](elem* buf, const T* src, long rstride, int vrows,
                         int c0, int cmax) {
-    if constexpr (BF) {
-      if (vec8) {
-        const int row = t >> 2, g = t & 3;
-        const int c = c0 + g * 8;
-        s16x8 val{};
-        if (row < vrows && c < cmax)
-          val = *reinterpret_cast<const s16x8*>(src + row * rstride + c);
-        TO::store_vec8(buf, row, g, val);
-        return;
-      }
-    }
-    for (int i = t; i < 64 * BK; i += 256) {
-      int row = i / BK, c = c0 + i % BK;
-      float v_ = 0.f;
-      if (row < vrows && c < cmax) v_ = to_f32(src[row * rstride + c]);
-      TO::store(buf, row, i % BK, v_);
-    }
+    stage_tile_rows<T>(t, buf, src, rstride, vrows, c0, cmax, vec8);
   };
   auto stage_trans = [&](elem* buf, const T* src, long rstride, int vrows,
                          int c0, int cmax) {
-    if constexpr (BF) {
-      if (vec8) {
-        for (int task = t; task < 512; task += 256) {
-          const int srow = task >> 3, g = task & 7;
-          const int c = c0 + g * 8;
-          s16x8 val{};
-          if (srow < vrows && c < cmax)
-            val = *reinterpret_cast<const s16x8*>(src + srow * rstride + c);
-          const u16* pv = reinterpret_cast<const u16*>(&val);
-          elem* tb = buf + (srow / BK) * 64 * ROW;
-          const int kk = srow % BK;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) tb[(g * 8 + j) * ROW + kk] = pv[j];
-        }
-        return;
-      }
-    }
-    for (int i = t; i < 64 * 64; i += 256) {
-      int srow = i >> 6, c = c0 + (i & 63);
-      float v_ = 0.f;
-      if (srow < vrows && c < cmax) v_ = to_f32(src[srow * rstride + c]);
-      TO::store(buf + (srow / BK) * 64 * ROW, i & 63, srow % BK, v_);
-    }
+    stage_tile_trans<T>(t, buf, src, rstride, vrows, c0, cmax, vec8);
   };
 
   // ---------- phase A: S_c = K_c Q^T -> A_c via saved (m, l) ----------

@@ -108,6 +108,72 @@ template <> struct TileOps<float> {
   }
 };
 
+// ---- shared MFMA-operand staging (attention kernels) ----
+// stage a [64 rows x BK] tile from row-major global memory: rows >=
+// vrows and columns >= cmax zero-filled. bf16 path moves 16-B s16x8
+// granules as raw bits; f32 falls back to scalar stores. Caller must
+// pass vec8 = (strides % 8 == 0) for the vector path. One call covers
+// exactly the 256-thread block's work; __syncthreads is the caller's.
+template <typename T>
+GFA_DEV void stage_tile_rows(int t, typename TileOps<T>::elem* buf,
+                             const T* src, long rstride, int vrows, int c0,
+                             int cmax, bool vec8) {
+  using TO = TileOps<T>;
+  constexpr int BK = TO::BK;
+  if constexpr (sizeof(typename TO::elem) == 2) {
+    if (vec8) {
+      const int row = t >> 2, g = t & 3;  // 256 tasks exactly
+      const int c = c0 + g * 8;
+      s16x8 val{};
+      if (row < vrows && c < cmax)
+        val = *reinterpret_cast<const s16x8*>(src + row * rstride + c);
+      TO::store_vec8(buf, row, g, val);
+      return;
+    }
+  }
+  for (int i = t; i < 64 * BK; i += 256) {
+    int row = i / BK, c = c0 + i % BK;
+    float v_ = 0.f;
+    if (row < vrows && c < cmax) v_ = to_f32(src[row * rstride + c]);
+    TO::store(buf, row, i % BK, v_);
+  }
+}
+
+// stage a transposed [64 global-cols x 64 src-rows] tile PAIR:
+// buf rows = global column index (c0 + 0..63), depth = source row
+// (two BK tiles back-to-back at stride 64*ROW).
+template <typename T>
+GFA_DEV void stage_tile_trans(int t, typename TileOps<T>::elem* buf,
+                              const T* src, long rstride, int vrows, int c0,
+                              int cmax, bool vec8) {
+  using TO = TileOps<T>;
+  constexpr int BK = TO::BK;
+  constexpr int ROW = TO::ROW;
+  if constexpr (sizeof(typename TO::elem) == 2) {
+    if (vec8) {
+      for (int task = t; task < 512; task += 256) {
+        const int srow = task >> 3, g = task & 7;
+        const int c = c0 + g * 8;
+        s16x8 val{};
+        if (srow < vrows && c < cmax)
+          val = *reinterpret_cast<const s16x8*>(src + srow * rstride + c);
+        const u16* pv = reinterpret_cast<const u16*>(&val);
+        auto* tb = buf + (srow / BK) * 64 * ROW;
+        const int kk = srow % BK;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) tb[(g * 8 + j) * ROW + kk] = pv[j];
+      }
+      return;
+    }
+  }
+  for (int i = t; i < 64 * 64; i += 256) {
+    int srow = i >> 6, c = c0 + (i & 63);
+    float v_ = 0.f;
+    if (srow < vrows && c < cmax) v_ = to_f32(src[srow * rstride + c]);
+    TO::store(buf + (srow / BK) * 64 * ROW, i & 63, srow % BK, v_);
+  }
+}
+
 // ---- grid helpers ----
 GFA_DEV long global_tid() {
   return (long)blockIdx.x * blockDim.x + threadIdx.x;

@@ -104,6 +104,10 @@ struct GemmParams {
 template <typename T>
 void launch_gemm_skinny(T*, const T*, const T*, const GemmParams&, bool,
                         hipStream_t);
+template <typename T>
+void launch_pack_w_blocked(T*, const T*, int, int, int, hipStream_t);
+template <typename T>
+void launch_pack_w_o9i(T*, const T*, int, int, int, hipStream_t);
 
 }  // namespace gfa
 
@@ -290,22 +294,21 @@ Tensor conv2d_fwd(Tensor x, Tensor w, int64_t stride, int64_t pad) {
   } else if (gfa::conv2d_slab_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
                                        p.kw, p.stride, p.pad, p.per_sample)) {
     // blocked weight layout [mT][cB][9][128][32] (o zero-padded to 128):
-    // one wave's A fragment set per (tap, cb) is a contiguous 1 KB read
+    // one wave's A fragment set per (tap, cb) is a contiguous 1 KB read;
+    // packed by one kernel launch (pack.hip), not a torch op chain
     int mT = (O + 127) / 128, cB = I / 32;
-    Tensor wsrc = w;
-    if (O % 128 != 0)
-      wsrc = torch::constant_pad_nd(w.reshape({O, (int64_t)I * 9}),
-                                    {0, 0, 0, (int64_t)mT * 128 - O})
-                 .reshape({(int64_t)mT * 128, I, 3, 3});
-    auto wb = wsrc.reshape({mT, 128, cB, 32, 9})
-                  .permute({0, 2, 4, 1, 3})
-                  .contiguous();
+    auto wb = torch::empty({mT, cB, 9, 128, 32}, w.options());
+    gfa::launch_pack_w_blocked<gfa::bf16>(ptr<gfa::bf16>(wb),
+                                          cptr<gfa::bf16>(w), O, I, mT,
+                                          cur_stream());
     gfa::launch_conv2d_fwd_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                      cptr<gfa::bf16>(wb), p.B, p.I, p.H, p.W,
                                      p.O, cur_stream());
   } else if (gfa::conv2d_s2_eligible(p.I, p.O, p.H, p.W, p.OH, p.OW, p.kh,
                                      p.kw, p.stride, p.pad, p.per_sample)) {
-    auto wr = w.permute({0, 2, 3, 1}).reshape({O, kh * kw, I}).contiguous();
+    auto wr = torch::empty({O, kh * kw, I}, w.options());
+    gfa::launch_pack_w_o9i<gfa::bf16>(ptr<gfa::bf16>(wr), cptr<gfa::bf16>(w),
+                                      O, I, kh * kw, cur_stream());
     gfa::launch_conv2d_s2_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                     cptr<gfa::bf16>(wr), p.B, p.I, p.H, p.W,
                                     p.O, cur_stream());
@@ -330,7 +333,9 @@ Tensor conv2d_up2(Tensor x, Tensor w) {
   TORCH_CHECK(w.size(1) == I && w.size(2) == 3 && w.size(3) == 3);
   TORCH_CHECK(gfa::conv2d_up2_eligible(I, O, H, W, 3, 3, 0),
               "conv2d_up2: shape not eligible (I%32, H%8, W%16)");
-  auto wr = w.permute({0, 2, 3, 1}).reshape({O, 9, I}).contiguous();
+  auto wr = torch::empty({O, 9, I}, w.options());
+  gfa::launch_pack_w_o9i<gfa::bf16>(ptr<gfa::bf16>(wr), cptr<gfa::bf16>(w),
+                                    O, I, 9, cur_stream());
   auto out = torch::empty({B, O, 2 * H, 2 * W}, x.options());
   gfa::launch_conv2d_up2_slab_bf16(ptr<gfa::bf16>(out), cptr<gfa::bf16>(x),
                                    cptr<gfa::bf16>(wr), B, I, H, W, O,

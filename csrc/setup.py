@@ -23,7 +23,8 @@ sources = [os.path.join(CSRC, f) for f in (
     "ext.hip", "fba.hip", "upfirdn2d.hip", "conv2d.hip", "conv2d_slab.hip",
     "conv2d_wgrad_slab.hip", "conv2d_up2.hip", "conv2d_s2.hip",
     "upfirdn2d_sep.hip", "modnorm.hip",
-    "mbstd.hip", "attn.hip", "attn_bwd.hip", "gemm_skinny.hip")]
+    "mbstd.hip", "attn.hip", "attn_bwd.hip", "gemm_skinny.hip",
+    "pack.hip")]
 
 setup(
     name="gansformer_amd_ext",
