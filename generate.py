@@ -30,6 +30,8 @@ def main(argv=None):
                    help="also save each image separately")
     p.add_argument("--attention-maps", action="store_true",
                    help="dump per-component attention heatmaps")
+    p.add_argument("--class", dest="class_idx", type=int, default=None,
+                   help="class index for conditional models (one-hot)")
     p.add_argument("--noise-mode", choices=["random", "const", "none"],
                    default="const")
     args = p.parse_args(argv)
@@ -42,6 +44,22 @@ def main(argv=None):
     Gs = Gs.to(device).eval()
     os.makedirs(args.output_dir, exist_ok=True)
 
+    label_dim = getattr(Gs, "label_dim", 0)
+    if label_dim > 0 and args.class_idx is None:
+        print(f"# conditional model (label_dim={label_dim}); sampling "
+              "random classes — pass --class N to fix one")
+
+    def make_label(n):
+        if label_dim == 0:
+            return None
+        lab = torch.zeros(n, label_dim, device=device)
+        if args.class_idx is not None:
+            lab[:, args.class_idx % label_dim] = 1.0
+        else:
+            idx = torch.randint(label_dim, (n,), device=device)
+            lab[torch.arange(n, device=device), idx] = 1.0
+        return lab
+
     torch.manual_seed(args.seed)
     imgs = []
     with torch.no_grad():
@@ -49,7 +67,8 @@ def main(argv=None):
         while remaining > 0:
             n = min(args.batch_size, remaining)
             z = Gs.sample_z(n, device=device)
-            img = Gs(z, truncation_psi=args.truncation_psi,
+            img = Gs(z, label=make_label(n),
+                     truncation_psi=args.truncation_psi,
                      noise_mode=args.noise_mode)
             imgs.append(img.cpu())
             remaining -= n
