@@ -25,7 +25,8 @@ def build_configs(args):
     dataset_args = EasyDict(
         dataset=args.dataset, data_dir=args.data_dir,
         resolution=args.res, synthetic_size=args.synthetic_size,
-        seed=args.seed, mirror_augment=args.mirror_augment)
+        seed=args.seed, mirror_augment=args.mirror_augment,
+        label_dim=args.label_dim)
     if args.inception_path:
         dataset_args.inception_path = args.inception_path
     G_args = EasyDict(
@@ -73,6 +74,10 @@ def main(argv=None):
     p.add_argument("--data-dir", default="datasets")
     p.add_argument("--res", "--resolution", dest="res", type=int, default=256)
     p.add_argument("--synthetic-size", type=int, default=50000)
+    p.add_argument("--label-dim", type=int, default=0,
+                   help="one-hot label dim for synthetic data (prepared "
+                        "datasets carry their own labels; >0 trains "
+                        "class-conditional G/D)")
     p.add_argument("--mirror-augment", action="store_true")
     # model
     p.add_argument("--transformer", choices=["none", "simplex", "duplex"],
