@@ -559,7 +559,10 @@ def test_slab_kernels_deterministic(dev):
     # that and nothing else).
     dy = torch.randn(2, 128, 64, 64, device=dev, dtype=torch.bfloat16)
     ref = _C().conv2d_wgrad(x2, dy, 1, 1, 3, 3, False).float()
+    # one bf16 output ulp at the observed magnitude (fp32 atomic-order
+    # noise rounds to at most +-1 ulp of the bf16 result)
+    tol = 2.0 ** -7 * ref.abs().max().item() + 1e-3
     for _ in range(8):
         out = _C().conv2d_wgrad(x2, dy, 1, 1, 3, 3, False).float()
         err = (out - ref).abs().max().item()
-        assert err <= 0.5, f"wgrad run-to-run drift {err} beyond ulp scale"
+        assert err <= tol, f"wgrad run-to-run drift {err} > {tol}"
