@@ -137,6 +137,25 @@ def bench_attn_bwd(dtype=torch.bfloat16):
               f"({te / tf:4.1f}x)")
 
 
+def bench_attn_bwd_one(dtype=torch.bfloat16):
+    """Single hot shape, many iters — for rocprofv3 runs (small-N bwd)."""
+    import math as _m
+    dev = "cuda:0"
+    B, Nq, Nk, D, E = 32, 4096, 17, 512, 512
+    scale = 1.0 / _m.sqrt(D)
+    q = torch.randn(B, Nq, D, device=dev, dtype=dtype)
+    k = torch.randn(B, Nk, D, device=dev, dtype=dtype)
+    v = torch.randn(B, Nk, E, device=dev, dtype=dtype)
+    dout = torch.randn(B, Nq, E, device=dev, dtype=dtype)
+    out, ml = C.bipartite_attn_fwd(q, k, v, scale)
+    drow = (dout.float() * out.float()).sum(-1).contiguous()
+    t = timeit(lambda: C.bipartite_attn_bwd(q, k, v, dout, drow, ml, scale),
+               iters=30, warmup=5)
+    fl = 5.0 * B * Nq * Nk * (D + E)
+    print(f"attnbwd1 smalln res64 d512: {t * 1e3:8.3f} ms "
+          f"({fl / t / 1e12:6.1f} TF)")
+
+
 def bench_gemm_skinny(dtype=torch.bfloat16):
     """Custom tall-skinny MFMA GEMM vs hipBLASLt (x @ w^T)."""
     dev = "cuda:0"
@@ -218,6 +237,8 @@ if __name__ == "__main__":
         bench_attn()
     if which in ("attnbwd", "all"):
         bench_attn_bwd()
+    if which == "attnbwd1":
+        bench_attn_bwd_one()
     if which in ("gemmsk", "all"):
         bench_gemm_skinny()
     if which in ("upfirdn", "all"):
