@@ -37,7 +37,7 @@ void attn_smalln_kernel(T* __restrict__ out, const T* __restrict__ q,
   __shared__ elem Ks[64 * ROW];
   __shared__ elem Ps[64 * (2 * ROW)];   // 64 q rows x 64 key cols (2 BK tiles)
   __shared__ elem Vt[64 * (2 * ROW)];   // 64 e rows x 64 key cols
-  __shared__ float Ss[64][68];          // raw scores
+  __shared__ float Ss[64][65];          // raw scores (stride 65 = 1 mod 64 banks: column-parallel access conflict-free)
 
   const int b = blockIdx.y;
   const int q0 = blockIdx.x * 64;
@@ -143,7 +143,7 @@ void attn_longn_partial(float* __restrict__ ws_m, float* __restrict__ ws_l,
   __shared__ elem Qs[64 * ROW];
   __shared__ elem Pt[64 * (2 * ROW)];   // [q][64 keys]
   __shared__ elem Vt[64 * (2 * ROW)];   // [e][64 keys]
-  __shared__ float SsT[64][68];         // [q][key] transposed scores
+  __shared__ float SsT[64][65];         // [q][key] transposed scores (65: see Ss)
 
   const int b = blockIdx.y;
   const int c = blockIdx.x;             // chunk

@@ -50,7 +50,7 @@ void attn_smalln_bwd(T* __restrict__ dq, float* __restrict__ dkw,
   constexpr int ROW = TO::ROW;
   using elem = typename TO::elem;
   constexpr bool BF = (sizeof(elem) == 2);
-  __shared__ float Ss[64][68];          // A, later overwritten with dS
+  __shared__ float Ss[64][65];          // A, later dS (stride 65 = 1 mod 64 banks: the 68 stride made every column-parallel access 4-way conflicted - 8.2 conflict cycles/LDS inst measured, gpurun_out/r02_attnpmc)
   __shared__ elem Ta[64 * 2 * ROW];     // staging pair (depth up to 64)
   __shared__ elem Tb[64 * 2 * ROW];
 
@@ -255,7 +255,7 @@ void attn_longn_bwd(float* __restrict__ dqw, T* __restrict__ dk,
   constexpr int ROW = TO::ROW;
   using elem = typename TO::elem;
   constexpr bool BF = (sizeof(elem) == 2);
-  __shared__ float Ss[64][68];          // A_c, later dS_c ([key][q])
+  __shared__ float Ss[64][65];          // A_c, later dS_c ([key][q]; stride 65, see small-N note)
   __shared__ elem Ta[64 * 2 * ROW];
   __shared__ elem Tb[64 * 2 * ROW];
 
