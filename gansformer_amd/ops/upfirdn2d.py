@@ -88,6 +88,10 @@ def _separable8(f):
     else None."""
     if f.shape != (4, 4):
         return None
+    # derived filters (e.g. the backward's f.flip) carry their
+    # factorization as an attribute: no host round-trip, no cache churn
+    if hasattr(f, "_gfa_sep8"):
+        return f._gfa_sep8
     ent = _sep_cache.get(id(f))
     if ent is not None and ent[0]() is f:
         return ent[1]
@@ -110,16 +114,26 @@ def _separable8(f):
     return res
 
 
+_MISSING = object()
+
+
 class _Upfirdn2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, f, up, dn, pad, gain):
         ctx.params = (up, dn, pad, gain, x.shape)
         ctx.save_for_backward(f)
+        # resolve the separable factorization ONCE per call graph:
+        # derived filters (backward flips) carry it as an attribute,
+        # stable buffers hit the id cache; ctx carries it to backward
+        # so the flip chain never re-checks on the host (a device sync).
+        f8 = getattr(f, "_gfa_sep8", _MISSING)
+        if f8 is _MISSING:
+            f8 = _separable8(f)
+        ctx.f8 = f8
         if native.use_native(x):
             if (x.dtype == torch.bfloat16 and up[0] == up[1]
                     and dn[0] == dn[1]
                     and (up[0], dn[0]) in ((1, 1), (2, 1), (1, 2))):
-                f8 = _separable8(f)
                 if f8 is not None:
                     return native.require_ext().upfirdn2d_sep(
                         x.contiguous(), f8, up[0], dn[0],
@@ -145,7 +159,16 @@ class _Upfirdn2d(torch.autograd.Function):
         qx0 = fw - 1 - pad[2]
         qy1 = (H - 1) * up[0] + fh - qy0 - Hout * dn[0]
         qx1 = (W - 1) * up[1] + fw - qx0 - Wout * dn[1]
-        dx = _Upfirdn2d.apply(dy, f.flip([0, 1]), dn, up, (qy0, qy1, qx0, qx1), gain)
+        ff = f.flip([0, 1])
+        # the flip is a fresh tensor every backward: derive its
+        # separable factorization from this call's resolved one
+        # (outer(fy,fx) flipped = outer(fy.flip, fx.flip)) instead of
+        # re-checking on the host — that check is a device sync, and
+        # per-backward syncs cost ~5% of the whole step
+        s8 = ctx.f8
+        ff._gfa_sep8 = None if s8 is None else torch.cat(
+            [s8[:4].flip(0), s8[4:].flip(0)])
+        dx = _Upfirdn2d.apply(dy, ff, dn, up, (qy0, qy1, qx0, qx1), gain)
         return dx, None, None, None, None, None
 
 
