@@ -309,3 +309,67 @@ def test_fid_feature_gather_two_ranks():
         assert f.shape == (12, 4)
         vals = set(np.unique(f).tolist())
         assert vals == {0.0, 1.0}, f"gather missing a rank's features: {vals}"
+
+
+# ---------------------------------------------------------------------------
+# Collective launch-order stress: 20 full trainer steps with per-rank
+# style-mixing randomness (different graph branches per rank) and lazy
+# reg. The immediate-launch bucket design requires identical hook order
+# on every rank; divergent params or a hang here would catch an
+# order/participation mismatch before it deadlocks real RCCL.
+# ---------------------------------------------------------------------------
+
+def _worker_stress(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    import copy
+    import torch.distributed as dist
+    from gansformer_amd.models.networks import Discriminator, Generator
+    from gansformer_amd.parallel.ddp import broadcast_params
+    from gansformer_amd.parallel.dist import setup_distributed
+    from gansformer_amd.training.trainer import GANTrainer
+
+    setup_distributed(backend="gloo")
+    torch.manual_seed(0)
+    G = Generator(z_dim=32, w_dim=32, img_resolution=16, num_components=4,
+                  transformer="duplex", channel_base=512, channel_max=64,
+                  bf16_res_count=0, mapping_layers=2, style_mixing_prob=0.9)
+    D = Discriminator(img_resolution=16, channel_base=512, channel_max=64,
+                      mbstd_group_size=2, bf16_res_count=0)
+    Gs = copy.deepcopy(G).eval()
+    for p in Gs.parameters():
+        p.requires_grad_(False)
+    broadcast_params(G)
+    broadcast_params(D)
+    broadcast_params(Gs)
+    tr = GANTrainer(G, D, Gs, torch.device("cpu"), 2, 2 * world,
+                    d_reg_interval=3, g_reg_interval=2, bucket_mb=0.1)
+    torch.manual_seed(7000 + rank)  # rank-divergent mixing draws + data
+    for step in range(20):
+        tr.step(lambda: torch.randn(2, 3, 16, 16), step, step * 2 * world)
+    vec = torch.cat([p.detach().reshape(-1) for p in G.parameters()]
+                    + [p.detach().reshape(-1) for p in D.parameters()])
+    q.put((rank, vec.numpy().copy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_trainer_style_mixing_order_stress():
+    world = 4
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_stress, args=(r, world, 29769, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, vec = q.get(timeout=400)
+        results[rank] = torch.tensor(vec)
+    for p in procs:
+        p.join(timeout=30)
+    for rank in range(1, world):
+        assert torch.equal(results[rank], results[0]), f"rank {rank}"
+    assert torch.isfinite(results[0]).all()
