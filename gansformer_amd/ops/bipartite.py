@@ -19,10 +19,14 @@ a later optimization.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 
 from . import native
+
+# debug escape hatch: force the eager recompute backward everywhere
+_FORCE_EAGER_BWD = os.environ.get("GFA_ATTN_BWD_EAGER") == "1"
 
 
 def _up(t):
@@ -69,7 +73,8 @@ class _BipartiteAttn(torch.autograd.Function):
         # fused kernel is correct but 0.7x (its 5-phase LDS dance can't
         # beat hipBLASLt's 400+ TF/s on the skinny recompute GEMMs), so
         # small-N keeps the eager composition.
-        if ctx.native and k.shape[1] > 64 and not torch.is_grad_enabled():
+        if (ctx.native and k.shape[1] > 64 and not _FORCE_EAGER_BWD
+                and not torch.is_grad_enabled()):
             # rebuild A from the saved softmax stats; drow uses the
             # flash identity rowsum(dA*A) == rowsum(dO*O).
             dout_c = dout.contiguous()
