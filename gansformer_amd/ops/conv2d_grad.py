@@ -137,8 +137,18 @@ def _unfold_batched(x, kh, kw, pad, stride):
     """F.unfold with the batch folded into channels: torch's im2col
     launches one kernel PER SAMPLE, which turns the small-conv GEMM
     routing into a launch storm; per-channel independence makes
-    unfold([1, B*I, H, W]) identical in one launch."""
+    unfold([1, B*I, H, W]) identical in one launch.
+
+    The fold is capped at B*I <= 16384 channels: beyond that the
+    ROCm im2col path memory-faults (bisected 2026-09-14: batch 64 x
+    512 ch = 32768 folded channels aborts with a write to a read-only
+    page inside the D backward; 56 x 512 = 28672 is clean — see
+    KNOWN_ISSUES.md for the full bisect). Above the cap we pay torch's
+    per-sample launches; this branch only serves the small-feature-map
+    tail, so the cost is bounded."""
     B, I, H, W = x.shape
+    if B * I > 16384:
+        return F.unfold(x, (kh, kw), padding=pad, stride=stride)
     u = F.unfold(x.reshape(1, B * I, H, W), (kh, kw), padding=pad,
                  stride=stride)
     return u.reshape(B, I * kh * kw, -1)
