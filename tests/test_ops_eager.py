@@ -315,3 +315,17 @@ def test_mbstd_analytic_backward_matches_autograd():
                                 params=(G, F, eps))
     dx = _MbStdStats.backward(ctx, dstats)[0]
     assert torch.allclose(dx, dx_ref, atol=1e-6, rtol=1e-5)
+
+
+def test_unfold_batched_fold_and_fallback_agree():
+    """_unfold_batched's folded path (batch into channels) and its
+    >16384-channel fallback (plain per-sample unfold) must be
+    identical; the cap exists because the folded ROCm im2col faults at
+    32768 channels (KNOWN_ISSUES.md)."""
+    import torch.nn.functional as F
+    from gansformer_amd.ops import conv2d_grad as cg
+    torch.manual_seed(0)
+    x = torch.randn(3, 5, 9, 9)
+    folded = cg._unfold_batched(x, 3, 3, 1, 2)
+    plain = F.unfold(x, (3, 3), padding=1, stride=2)
+    assert torch.equal(folded, plain)
