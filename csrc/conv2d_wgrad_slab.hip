@@ -302,7 +302,12 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
     }
   };
 
-  // copy s, dst j: x column 2*(col0+j)+s-1 = window[2j+s+3-32h]
+  // copy s, dst j: x column 2*(col0+j)+s-1 = window[2j+s+3-32h], and
+  // with j0 = 16*s_h the index 2*(j0+j)+s+3-32*s_h folds to 2j+s+3 —
+  // written in the folded form so every tk access is a compile-time
+  // constant and tk stays in registers (the unfolded runtime-s_h form
+  // forced tk to 96 B/lane of scratch; see KNOWN_ISSUES on the
+  // batch-64 scratch-fault suspect)
   auto win_write = [&](int sb) {
     if (!stager) return;
     const u16* rp = reinterpret_cast<const u16*>(&tk[0]);
@@ -313,7 +318,7 @@ void conv2d_wgrad_slab_s2_bf16(float* __restrict__ ws,  // [O][9][I] f32
       u16* dst = base + s * SLABX;
 #pragma unroll
       for (int j = 0; j < 16; ++j)
-        dst[j0 + j] = rp[2 * (j0 + j) + s + 3 - 32 * s_h];
+        dst[j0 + j] = rp[2 * j + s + 3];
     }
   };
 
