@@ -96,3 +96,24 @@ def test_mirror_augment_dataset():
     img_f, _ = ds[3 + 8]
     assert torch.equal(img_f, torch.flip(img, dims=[-1]))
     assert not torch.equal(img_f, img)
+
+
+@pytest.mark.timeout(300)
+def test_conditional_train_and_generate_cli(tmp_path):
+    """--label-dim trains a conditional model end to end; generate.py
+    detects label_dim from the pkl and honors --class."""
+    r = run(["train.py", "--result-dir", str(tmp_path / "runs"),
+             "--dataset", "synthetic", "--res", "16", "--label-dim", "3",
+             "--transformer", "simplex", "--components-num", "2",
+             "--latent-size", "16", "--dlatent-size", "16",
+             "--fmap-base", "256", "--fmap-max", "16",
+             "--mapping-layers", "1", "--bf16-res", "0",
+             "--batch-gpu", "2", "--total-kimg", "0.002",
+             "--snapshot-kimg", "0.002", "--synthetic-size", "8",
+             "--num-workers", "0", "--mbstd-group", "2"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    pkl = glob.glob(str(tmp_path / "runs" / "*" / "network-snapshot-*.pkl"))[0]
+    r = run(["generate.py", "--network", pkl, "--num-images", "2",
+             "--class", "1", "--output-dir", str(tmp_path / "gen")])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(tmp_path / "gen" / "grid.png")
