@@ -28,7 +28,7 @@ from __future__ import annotations
 import torch
 
 from .conv2d_grad import conv2d_gradfix, conv2d_up2
-from .upfirdn2d import upsample2d, upfirdn2d
+from .upfirdn2d import upfirdn2d
 
 
 def modulated_conv2d(
