@@ -73,6 +73,22 @@ def barrier():
         dist.barrier()
 
 
+def any_rank(flag: bool) -> bool:
+    """Collective OR of a per-rank boolean. Used for stop decisions:
+    abort files / SIGTERM may reach only one rank, and a rank leaving
+    the training loop alone would hang the others on the next
+    collective."""
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return flag
+    backend = str(dist.get_backend()).lower()
+    dev = torch.device("cuda", torch.cuda.current_device()) \
+        if backend in ("nccl", "rccl") and torch.cuda.is_available() \
+        else torch.device("cpu")
+    t = torch.tensor([1 if flag else 0], dtype=torch.int32, device=dev)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return bool(t.item())
+
+
 def cleanup():
     if dist.is_initialized():
         dist.destroy_process_group()

@@ -24,7 +24,7 @@ from ..metrics.fid import METRIC_SPECS, compute_fid, load_extractor
 from ..metrics.scalars import ScalarLogger
 from ..models.networks import Discriminator, Generator
 from ..parallel import broadcast_params
-from ..parallel.dist import barrier, cleanup, setup_distributed
+from ..parallel.dist import any_rank, barrier, cleanup, setup_distributed
 from ..rundir import RunContext
 from . import snapshot as snap
 from .dataset import load_dataset, make_loader, normalize_images
@@ -252,7 +252,10 @@ def training_loop(
                 fakes, os.path.join(run_dir, f"fakes{cur_nimg // 1000:06d}.png"))
             next_img = cur_nimg + image_snapshot_kimg * 1000
 
-        want_stop = stop_flag["stop"] or ctx.should_stop()
+        # stop decisions must be COLLECTIVE: the abort file exists only
+        # on rank 0 and SIGTERM may reach a subset of ranks; a rank
+        # leaving alone would hang the others on the next collective
+        want_stop = any_rank(stop_flag["stop"] or ctx.should_stop())
         if cur_nimg >= next_snap or cur_nimg >= total_kimg * 1000 or want_stop:
             trainer.sync_ranks()
             if is_main and run_dir is not None:

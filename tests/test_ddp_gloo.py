@@ -373,3 +373,57 @@ def test_trainer_style_mixing_order_stress():
     for rank in range(1, world):
         assert torch.equal(results[rank], results[0]), f"rank {rank}"
     assert torch.isfinite(results[0]).all()
+
+
+# ---------------------------------------------------------------------------
+# Collective stop: an abort signal seen by ONE rank must stop ALL ranks
+# cleanly (rank-local want_stop would hang the others on the next
+# collective).
+# ---------------------------------------------------------------------------
+
+def _worker_stop(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    import torch.distributed as dist
+    from gansformer_amd.training.loop import training_loop
+    import tempfile
+    run_dir = os.path.join(tempfile.gettempdir(), f"stoptest-{port}")
+    if rank == 0:
+        os.makedirs(run_dir, exist_ok=True)
+        with open(os.path.join(run_dir, "abort.txt"), "w") as f:
+            f.write("stop")
+    out = training_loop(
+        run_dir=run_dir if rank == 0 else None,
+        dataset_args=dict(dataset="synthetic", resolution=16,
+                          synthetic_size=16),
+        G_args=dict(z_dim=16, w_dim=16, num_components=2,
+                    transformer="none", channel_base=256, channel_max=16,
+                    bf16_res_count=0, mapping_layers=1),
+        D_args=dict(channel_base=256, channel_max=16, mbstd_group_size=2,
+                    bf16_res_count=0),
+        total_kimg=10.0,  # would be ~2500 steps without the abort
+        batch_gpu=2, snapshot_kimg=5.0, image_snapshot_kimg=5.0,
+        log_interval_kimg=0.004, num_workers=0, seed=0)
+    q.put((rank, out["steps"]))
+
+
+@pytest.mark.timeout(300)
+def test_abort_stops_all_ranks():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_stop, args=(r, world, 29770, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    steps = {}
+    for _ in range(world):
+        rank, s = q.get(timeout=280)
+        steps[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    # both ranks stopped early (well before 2500 steps) at the same step
+    assert steps[0] == steps[1]
+    assert steps[0] < 50, steps
