@@ -181,3 +181,37 @@ def test_conditional_pkl_roundtrip(tmp_path):
         la = D(a, label=lab)
         lb = D2(a, label=lab)
     assert torch.allclose(la, lb, atol=1e-6)
+
+
+def test_tf_name_map_remaining_leaves():
+    """ToRGB / Conv(4x4) / noise_strength / dlatent_avg broadcast."""
+    import numpy as np
+    from gansformer_amd.models.networks import Generator
+    from gansformer_amd.pkl_compat import load_variables
+    G = Generator(z_dim=32, w_dim=32, img_resolution=16, num_components=4,
+                  transformer="none", channel_base=512, channel_max=64,
+                  bf16_res_count=0, mapping_layers=2)
+    sd = {k: v.clone() for k, v in G.state_dict().items()}
+    w = sd["synthesis.blocks.0.torgb.weight"].numpy()  # [3, C, 1, 1]
+    tf_vars = [
+        ("G_synthesis/4x4/ToRGB/weight",
+         np.transpose(w, (2, 3, 1, 0)) + 1.0),
+        ("G_synthesis/4x4/ToRGB/mod_bias",
+         sd["synthesis.blocks.0.torgb.affine.bias"].numpy() + 1.0),
+        ("G_synthesis/4x4/Conv/noise_strength",
+         sd["synthesis.blocks.0.conv1.noise_strength"].numpy().reshape(())
+         + 1.0),
+        ("dlatent_avg", np.full((32,), 0.25, dtype=np.float32)),
+    ]
+    assert load_variables(G, tf_vars, strict=True) == []
+    sd2 = G.state_dict()
+    assert torch.allclose(sd2["synthesis.blocks.0.torgb.weight"],
+                          sd["synthesis.blocks.0.torgb.weight"] + 1.0)
+    assert torch.allclose(sd2["synthesis.blocks.0.torgb.affine.bias"],
+                          sd["synthesis.blocks.0.torgb.affine.bias"] + 1.0)
+    assert torch.allclose(
+        sd2["synthesis.blocks.0.conv1.noise_strength"],
+        sd["synthesis.blocks.0.conv1.noise_strength"] + 1.0)
+    # dlatent_avg broadcasts over all k+1 latents
+    assert torch.allclose(sd2["mapping.w_avg"],
+                          torch.full_like(sd["mapping.w_avg"], 0.25))
