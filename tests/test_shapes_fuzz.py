@@ -111,3 +111,45 @@ def test_mod_bias_act_fuzz(b, c, h, use_noise, use_bias):
     ref = torch.clamp(torch.nn.functional.leaky_relu(v, 0.2)
                       * _m.sqrt(2.0), -8.0, 8.0)
     assert torch.allclose(y, ref, atol=1e-10)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    b=st.integers(1, 3), nq=st.integers(1, 70), nk=st.integers(1, 70),
+    d=st.sampled_from([8, 16, 33]), e=st.sampled_from([8, 24]),
+)
+def test_bipartite_attention_fuzz(b, nq, nk, d, e):
+    """Attention op vs plain softmax composition over random shapes
+    (fwd + grads, fp64 so the CPU eager path is exact)."""
+    from gansformer_amd.ops.bipartite import bipartite_attention
+    torch.manual_seed(0)
+    q = torch.randn(b, nq, d, dtype=torch.float64, requires_grad=True)
+    k = torch.randn(b, nk, d, dtype=torch.float64, requires_grad=True)
+    v = torch.randn(b, nk, e, dtype=torch.float64, requires_grad=True)
+    out = bipartite_attention(q, k, v)
+    a = torch.softmax((q @ k.transpose(1, 2)) * d ** -0.5, dim=-1)
+    ref = a @ v
+    assert torch.allclose(out, ref, atol=1e-10)
+    g = torch.randn_like(out)
+    got = torch.autograd.grad(out, (q, k, v), g, retain_graph=True)
+    want = torch.autograd.grad(ref, (q, k, v), g)
+    for gg, ww in zip(got, want):
+        assert torch.allclose(gg, ww, atol=1e-9)
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    b=st.integers(2, 6), c=st.integers(2, 8), hw=st.sampled_from([3, 5]),
+    gs=st.integers(1, 6), f=st.sampled_from([1, 2]),
+)
+def test_mbstd_fuzz(b, c, hw, gs, f):
+    from gansformer_amd.ops import minibatch_stddev
+    if c % f:
+        return
+    torch.manual_seed(1)
+    x = torch.randn(b, c, hw, hw, dtype=torch.float64, requires_grad=True)
+    y = minibatch_stddev(x, group_size=gs, num_channels=f)
+    assert y.shape == (b, c + f, hw, hw)
+    assert torch.allclose(y[:, :c], x)
+    y.square().sum().backward()
+    assert torch.isfinite(x.grad).all()
