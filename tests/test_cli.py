@@ -117,3 +117,27 @@ def test_conditional_train_and_generate_cli(tmp_path):
              "--class", "1", "--output-dir", str(tmp_path / "gen")])
     assert r.returncode == 0, r.stderr[-2000:]
     assert os.path.exists(tmp_path / "gen" / "grid.png")
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_world2(tmp_path):
+    """The driver's SCALE entry path: torchrun --nproc-per-node 2
+    bench.py ... must emit exactly one JSON line from rank 0 with the
+    whole-job aggregate (gloo on CPU here; RCCL on GPU boxes)."""
+    import json as _json
+    import subprocess, sys
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29882", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--res", "32",
+         "--batch-gpu", "2", "--components-num", "2"],
+        capture_output=True, text=True, timeout=560, cwd=str(tmp_path))
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    row = _json.loads(lines[0])
+    assert row["n_gpus"] == 2
+    assert row["config"]["global_batch"] == 4
+    assert row["config"]["parallelism"] == "dp2"
+    assert row["value"] > 0 and row["ms_per_step"] > 0
