@@ -427,3 +427,60 @@ def test_abort_stops_all_ranks():
     # both ranks stopped early (well before 2500 steps) at the same step
     assert steps[0] == steps[1]
     assert steps[0] < 50, steps
+
+
+# ---------------------------------------------------------------------------
+# Full training_loop with FID metrics on 2 ranks: the metric path's
+# feature gather and the snapshot-time sync_ranks must line up as
+# collectives on every rank (this is the path the 8-GPU run exercises).
+# ---------------------------------------------------------------------------
+
+def _worker_loop_metrics(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    import tempfile
+    from gansformer_amd.training.loop import training_loop
+    run_dir = None
+    if rank == 0:
+        run_dir = os.path.join(tempfile.gettempdir(), f"loopm-{port}")
+        os.makedirs(run_dir, exist_ok=True)
+    out = training_loop(
+        run_dir=run_dir,
+        dataset_args=dict(dataset="clevr-synth", resolution=16,
+                          synthetic_size=32),
+        G_args=dict(z_dim=16, w_dim=16, num_components=2,
+                    transformer="simplex", channel_base=256, channel_max=16,
+                    bf16_res_count=0, mapping_layers=1),
+        D_args=dict(channel_base=256, channel_max=16, mbstd_group_size=2,
+                    bf16_res_count=0),
+        total_kimg=0.008, batch_gpu=2,  # 4 global/step -> 2 steps
+        metrics=["fid256"], metric_images=8, metric_kimg=0.008,
+        snapshot_kimg=0.008, image_snapshot_kimg=0.016,
+        log_interval_kimg=0.004, num_workers=0, seed=0)
+    fid_file = (os.path.join(run_dir, "metric-fid256.txt")
+                if rank == 0 else None)
+    n_fid = 0
+    if fid_file and os.path.exists(fid_file):
+        n_fid = len(open(fid_file).readlines())
+    q.put((rank, out["steps"], n_fid))
+
+
+@pytest.mark.timeout(420)
+def test_training_loop_metrics_world2():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_loop_metrics,
+                         args=(r, world, 29771, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        rank, steps, n_fid = q.get(timeout=400)
+        res[rank] = (steps, n_fid)
+    for p in procs:
+        p.join(timeout=30)
+    assert res[0][0] == res[1][0] == 2
+    assert res[0][1] >= 1  # FID eval recorded by rank 0
