@@ -199,6 +199,11 @@ def training_loop(
 
     cur_nimg = start_nimg
     step = start_nimg // batch_size
+    # SIGTERM/abort reaction latency: every ~quarter log interval, at
+    # least every 16 steps' worth of images
+    stopchk_nimg = max(batch_size, min(log_interval_kimg * 250,
+                                       batch_size * 16))
+    next_stopchk = cur_nimg + stopchk_nimg
     next_log = cur_nimg + log_interval_kimg * 1000
     next_snap = cur_nimg + snapshot_kimg * 1000
     next_img = 0  # write fakes at start too
@@ -252,10 +257,17 @@ def training_loop(
                 fakes, os.path.join(run_dir, f"fakes{cur_nimg // 1000:06d}.png"))
             next_img = cur_nimg + image_snapshot_kimg * 1000
 
-        # stop decisions must be COLLECTIVE: the abort file exists only
+        # Stop decisions must be COLLECTIVE: the abort file exists only
         # on rank 0 and SIGTERM may reach a subset of ranks; a rank
-        # leaving alone would hang the others on the next collective
-        want_stop = any_rank(stop_flag["stop"] or ctx.should_stop())
+        # leaving alone would hang the others on the next collective.
+        # The collective runs on the lockstep stop-check cadence (not
+        # every step: the 4-byte all-reduce + .item() is a cross-rank
+        # hard sync that would cap backward/comm overlap), so stop
+        # latency is bounded by stopchk_interval.
+        want_stop = False
+        if cur_nimg >= next_stopchk or cur_nimg >= total_kimg * 1000:
+            want_stop = any_rank(stop_flag["stop"] or ctx.should_stop())
+            next_stopchk = cur_nimg + stopchk_nimg
         if cur_nimg >= next_snap or cur_nimg >= total_kimg * 1000 or want_stop:
             trainer.sync_ranks()
             if is_main and run_dir is not None:
