@@ -95,6 +95,14 @@ class _Conv2dWgrad(torch.autograd.Function):
     def forward(ctx, x, dy, stride, pad, kh, kw, per_sample):
         ctx.save_for_backward(x, dy)
         ctx.params = (stride, pad, kh, kw, per_sample)
+        # NOTE on fp32 operands below: every faulting variant of the
+        # batch-64 D-backward abort (KNOWN_ISSUES.md) contained one of
+        # these double-batched bf16 bmm calls, and every variant that
+        # replaced exactly them with fp32 passed — the bf16 strided
+        # batched hipBLASLt path at batch 64 is the prime suspect.
+        # fp32 here also matches the slab wgrad's accumulation
+        # precision, and these branches only serve small maps, so the
+        # cost is negligible.
         if native.use_native(x, dy):
             if _is_1x1(kh, kw, stride, pad) and not per_sample:
                 # 1x1 wgrad is a plain GEMM: dw = sum_b dY_b X_b^T
@@ -102,17 +110,18 @@ class _Conv2dWgrad(torch.autograd.Function):
                 # skinny tRGB/fromRGB shapes)
                 B, I = x.shape[0], x.shape[1]
                 O = dy.shape[1]
-                dw = torch.matmul(dy.reshape(B, O, -1),
-                                  x.reshape(B, I, -1).transpose(1, 2))
-                return dw.sum(0).reshape(O, I, 1, 1)
+                dw = torch.matmul(dy.reshape(B, O, -1).float(),
+                                  x.reshape(B, I, -1).transpose(1, 2)
+                                  .float())
+                return dw.sum(0).reshape(O, I, 1, 1).to(x.dtype)
             if not per_sample and dy.shape[2] * dy.shape[3] <= 256:
                 # small feature maps: im2col + batched GEMM (see fwd)
                 B, I = x.shape[0], x.shape[1]
                 O = dy.shape[1]
                 unf = _unfold_batched(x, kh, kw, pad, stride)
-                dw = torch.matmul(dy.reshape(B, O, -1),
-                                  unf.transpose(1, 2))
-                return dw.sum(0).reshape(O, I, kh, kw)
+                dw = torch.matmul(dy.reshape(B, O, -1).float(),
+                                  unf.transpose(1, 2).float())
+                return dw.sum(0).reshape(O, I, kh, kw).to(x.dtype)
             return native.require_ext().conv2d_wgrad(
                 x.contiguous(), dy.contiguous(), stride, pad, kh, kw, per_sample)
         return _eager_wgrad(x, dy, stride, pad, kh, kw, per_sample)
